@@ -1,0 +1,162 @@
+#include "mttkrp_cpu.hpp"
+#include <cstring>
+#if defined(_OPENMP)
+#include <omp.h>
+#endif
+
+namespace splatt {
+
+template <typename V>
+void mttkrp_stream(const SpTensor<V> & tt, V const * const * mats,
+                   V * out, int mode, int rank) {
+  const int nm = tt.nmodes;
+  std::memset(out, 0, sizeof(V) * tt.dims[mode] * rank);
+  std::vector<V> accum(rank);
+  for (idx_t x = 0; x < tt.nnz; ++x) {
+    const V v = tt.vals[x];
+    for (int f = 0; f < rank; ++f) accum[f] = v;
+    for (int m = 0; m < nm; ++m) {
+      if (m == mode) continue;
+      const V * row = mats[m] + tt.ind[m][x] * rank;
+      for (int f = 0; f < rank; ++f) accum[f] *= row[f];
+    }
+    V * orow = out + tt.ind[mode][x] * rank;
+    for (int f = 0; f < rank; ++f) orow[f] += accum[f];
+  }
+}
+
+namespace {
+
+// Subtree accumulation for levels (level..leaf): returns into `buf` the
+// F-vector g(node) = M_l[fid] (.) sum_children g(child), leaf g = v * M_leaf[fid].
+// `include_self`: whether to multiply by this node's own factor row.
+template <typename V>
+struct CsfWalker {
+  const Csf<V> & c;
+  V const * const * mats;   // indexed by MODE
+  V * out;
+  int rank;
+  int outdepth;
+  int leaf;
+
+  // accumulate g over children of (level, node) into acc (acc zeroed by caller)
+  void subtree_below(int level, int64_t node, V * acc, V * scratch) const {
+    const int child_level = level + 1;
+    const int64_t start = c.fptr[level][node];
+    const int64_t end = c.fptr[level][node + 1];
+    const V * M = mats[c.dim_perm[child_level]];
+    if (child_level == leaf) {
+      for (int64_t j = start; j < end; ++j) {
+        const V v = c.vals[j];
+        const V * row = M + (idx_t)c.fids[leaf][j] * rank;
+        for (int f = 0; f < rank; ++f) acc[f] += v * row[f];
+      }
+    } else {
+      for (int64_t n = start; n < end; ++n) {
+        for (int f = 0; f < rank; ++f) scratch[f] = 0;
+        subtree_below(child_level, n, scratch, scratch + rank);
+        const V * row = M + (idx_t)c.fids[child_level][n] * rank;
+        for (int f = 0; f < rank; ++f) acc[f] += scratch[f] * row[f];
+      }
+    }
+  }
+
+  // walk down maintaining the Hadamard product of ancestor rows (`above`),
+  // emit at outdepth (intl/leaf output classes)
+  void walk_down(int level, int64_t node, const V * above, V * bufs) const {
+    const int64_t start = c.fptr[level][node];
+    const int64_t end = c.fptr[level][node + 1];
+    const int child_level = level + 1;
+    const V * M = mats[c.dim_perm[child_level]];
+    if (child_level == outdepth) {
+      if (child_level == leaf) {
+        // leaf output: out[fid] += v * above
+        for (int64_t j = start; j < end; ++j) {
+          const V v = c.vals[j];
+          V * orow = out + (idx_t)c.fids[leaf][j] * rank;
+          for (int f = 0; f < rank; ++f) {
+            #pragma omp atomic
+            orow[f] += v * above[f];
+          }
+        }
+      } else {
+        // internal output: out[fid] += above (.) below
+        for (int64_t n = start; n < end; ++n) {
+          V * below = bufs;
+          for (int f = 0; f < rank; ++f) below[f] = 0;
+          subtree_below(child_level, n, below, bufs + rank);
+          V * orow = out + (idx_t)c.fids[child_level][n] * rank;
+          for (int f = 0; f < rank; ++f) {
+            #pragma omp atomic
+            orow[f] += above[f] * below[f];
+          }
+        }
+      }
+    } else {
+      for (int64_t n = start; n < end; ++n) {
+        V * nxt = bufs;
+        const V * row = M + (idx_t)c.fids[child_level][n] * rank;
+        for (int f = 0; f < rank; ++f) nxt[f] = above[f] * row[f];
+        walk_down(child_level, n, nxt, bufs + rank);
+      }
+    }
+  }
+};
+
+}  // namespace
+
+template <typename V>
+void mttkrp_csf_cpu(const Csf<V> & c, V const * const * mats,
+                    V * out, int mode, int rank, int nthreads) {
+  const int nm = c.nmodes;
+  const int outdepth = c.level_of_mode(mode);
+  const int leaf = nm - 1;
+  std::memset(out, 0, sizeof(V) * c.dims[mode] * rank);
+  if (c.nnz == 0) return;
+
+  const bool root_labeled = !c.fids[0].empty();
+  const int64_t nroot = (int64_t)c.nfibs[0];
+
+#if defined(_OPENMP)
+  if (nthreads > 0) omp_set_num_threads(nthreads);
+#endif
+
+  CsfWalker<V> w{c, mats, out, rank, outdepth, leaf};
+
+  if (outdepth == 0) {
+    // root output: no write conflicts across root nodes
+    #pragma omp parallel
+    {
+      std::vector<V> bufs((size_t)rank * (nm + 1));
+      #pragma omp for schedule(dynamic, 16)
+      for (int64_t s = 0; s < nroot; ++s) {
+        V * acc = bufs.data();
+        for (int f = 0; f < rank; ++f) acc[f] = 0;
+        w.subtree_below(0, s, acc, bufs.data() + rank);
+        const idx_t orow_i = root_labeled ? (idx_t)c.fids[0][s] : (idx_t)s;
+        V * orow = out + orow_i * rank;
+        for (int f = 0; f < rank; ++f) orow[f] += acc[f];
+      }
+    }
+  } else {
+    const V * Mroot = mats[c.dim_perm[0]];
+    #pragma omp parallel
+    {
+      std::vector<V> bufs((size_t)rank * (nm + 2));
+      #pragma omp for schedule(dynamic, 16)
+      for (int64_t s = 0; s < nroot; ++s) {
+        const idx_t rid = root_labeled ? (idx_t)c.fids[0][s] : (idx_t)s;
+        const V * above0 = Mroot + rid * rank;
+        if (outdepth == 0) continue;  // unreachable
+        w.walk_down(0, s, above0, bufs.data());
+      }
+    }
+  }
+}
+
+template void mttkrp_stream<float>(const SpTensor<float>&, float const* const*, float*, int, int);
+template void mttkrp_stream<double>(const SpTensor<double>&, double const* const*, double*, int, int);
+template void mttkrp_csf_cpu<float>(const Csf<float>&, float const* const*, float*, int, int, int);
+template void mttkrp_csf_cpu<double>(const Csf<double>&, double const* const*, double*, int, int, int);
+
+}  // namespace splatt

@@ -1,0 +1,446 @@
+// Python boundary of the MI355X-native sparse tensor engine.
+// CPU core functions exposed over torch tensors; HIP kernel launchers are
+// declared extern "C" (defined in csrc/hip/*.hip, compiled by hipcc for
+// gfx950 and linked in) and take raw device pointers + the current stream,
+// so this translation unit needs no HIP headers.
+#include <torch/extension.h>
+#include <vector>
+#include <string>
+
+#include "core/types.hpp"
+#include "core/sptensor.hpp"
+#include "core/csf.hpp"
+#include "core/mttkrp_cpu.hpp"
+#include "core/matrix.hpp"
+#include "core/cpd.hpp"
+#include "core/io.hpp"
+
+namespace sp = splatt;
+using torch::Tensor;
+
+// ---------------------------------------------------------------- helpers
+
+static sp::Options opts_from_dict(const py::dict & d) {
+  sp::Options o;
+  if (d.contains("tolerance")) o.tolerance = d["tolerance"].cast<double>();
+  if (d.contains("max_iters")) o.max_iters = d["max_iters"].cast<uint64_t>();
+  if (d.contains("seed")) o.seed = d["seed"].cast<uint64_t>();
+  if (d.contains("nthreads")) o.nthreads = d["nthreads"].cast<int>();
+  if (d.contains("csf_alloc")) {
+    const std::string a = d["csf_alloc"].cast<std::string>();
+    o.csf_alloc = a == "one" ? sp::CsfAlloc::ONEMODE
+                : a == "all" ? sp::CsfAlloc::ALLMODE
+                             : sp::CsfAlloc::TWOMODE;
+  }
+  return o;
+}
+
+template <typename V>
+static sp::SpTensor<V> coo_from_torch(const Tensor & inds, const Tensor & vals,
+                                      const std::vector<int64_t> & dims) {
+  TORCH_CHECK(inds.dim() == 2, "inds must be [nmodes, nnz]");
+  TORCH_CHECK(inds.scalar_type() == torch::kInt64, "inds must be int64");
+  auto ic = inds.contiguous();
+  auto vc = vals.contiguous();
+  const int nm = (int)ic.size(0);
+  const sp::idx_t nnz = (sp::idx_t)ic.size(1);
+  sp::idx_t d[sp::MAX_NMODES];
+  for (int m = 0; m < nm; ++m) d[m] = (sp::idx_t)dims[m];
+  sp::SpTensor<V> tt(nm, nnz, d);
+  const int64_t * ip = ic.data_ptr<int64_t>();
+  const V * vp = vc.data_ptr<V>();
+  for (int m = 0; m < nm; ++m)
+    for (sp::idx_t i = 0; i < nnz; ++i) tt.ind[m][i] = (sp::idx_t)ip[m * nnz + i];
+  std::copy(vp, vp + nnz, tt.vals.begin());
+  return tt;
+}
+
+template <typename V>
+static std::tuple<Tensor, Tensor, std::vector<int64_t>>
+coo_to_torch(const sp::SpTensor<V> & tt) {
+  const auto vopt = torch::TensorOptions().dtype(
+      std::is_same<V, double>::value ? torch::kFloat64 : torch::kFloat32);
+  Tensor inds = torch::empty({tt.nmodes, (int64_t)tt.nnz}, torch::kInt64);
+  Tensor vals = torch::empty({(int64_t)tt.nnz}, vopt);
+  int64_t * ip = inds.data_ptr<int64_t>();
+  V * vp = vals.data_ptr<V>();
+  for (int m = 0; m < tt.nmodes; ++m)
+    for (sp::idx_t i = 0; i < tt.nnz; ++i) ip[m * tt.nnz + i] = (int64_t)tt.ind[m][i];
+  std::copy(tt.vals.begin(), tt.vals.end(), vp);
+  std::vector<int64_t> dims(tt.nmodes);
+  for (int m = 0; m < tt.nmodes; ++m) dims[m] = (int64_t)tt.dims[m];
+  return {inds, vals, dims};
+}
+
+// csf as a python dict of torch tensors
+template <typename V>
+static py::dict csf_to_py(const sp::Csf<V> & c) {
+  py::dict d;
+  const auto vopt = torch::TensorOptions().dtype(
+      std::is_same<V, double>::value ? torch::kFloat64 : torch::kFloat32);
+  py::list fptr, fids;
+  for (int l = 0; l < c.nmodes; ++l) {
+    if (l < c.nmodes - 1) {
+      Tensor t = torch::empty({(int64_t)c.fptr[l].size()}, torch::kInt64);
+      std::copy(c.fptr[l].begin(), c.fptr[l].end(), t.data_ptr<int64_t>());
+      fptr.append(t);
+    } else {
+      fptr.append(py::none());
+    }
+    if (c.fids[l].empty()) {
+      fids.append(py::none());
+    } else {
+      Tensor t = torch::empty({(int64_t)c.fids[l].size()}, torch::kInt32);
+      const sp::fid_t * s = c.fids[l].data();
+      int32_t * p = t.data_ptr<int32_t>();
+      for (size_t i = 0; i < c.fids[l].size(); ++i) p[i] = (int32_t)s[i];
+      fids.append(t);
+    }
+  }
+  Tensor vals = torch::empty({(int64_t)c.vals.size()}, vopt);
+  std::copy(c.vals.begin(), c.vals.end(), vals.data_ptr<V>());
+  d["fptr"] = fptr;
+  d["fids"] = fids;
+  d["vals"] = vals;
+  std::vector<int64_t> dims(c.nmodes), perm(c.nmodes), nfibs(c.nmodes);
+  for (int l = 0; l < c.nmodes; ++l) {
+    dims[l] = (int64_t)c.dims[l];
+    perm[l] = c.dim_perm[l];
+    nfibs[l] = (int64_t)c.nfibs[l];
+  }
+  d["dims"] = dims;
+  d["dim_perm"] = perm;
+  d["nfibs"] = nfibs;
+  return d;
+}
+
+template <typename V>
+static sp::Csf<V> csf_from_py(const py::dict & d) {
+  sp::Csf<V> c;
+  auto dims = d["dims"].cast<std::vector<int64_t>>();
+  auto perm = d["dim_perm"].cast<std::vector<int64_t>>();
+  c.nmodes = (int)dims.size();
+  for (int l = 0; l < c.nmodes; ++l) {
+    c.dims[l] = (sp::idx_t)dims[l];
+    c.dim_perm[l] = (int)perm[l];
+    c.dim_iperm[perm[l]] = l;
+  }
+  py::list fptr = d["fptr"], fids = d["fids"];
+  for (int l = 0; l < c.nmodes; ++l) {
+    if (!fptr[l].is_none()) {
+      Tensor t = fptr[l].cast<Tensor>().contiguous();
+      c.fptr[l].assign(t.data_ptr<int64_t>(), t.data_ptr<int64_t>() + t.numel());
+    }
+    if (!fids[l].is_none()) {
+      Tensor t = fids[l].cast<Tensor>().contiguous();
+      const int32_t * p = t.data_ptr<int32_t>();
+      c.fids[l].resize(t.numel());
+      for (int64_t i = 0; i < t.numel(); ++i) c.fids[l][i] = (sp::fid_t)p[i];
+    }
+  }
+  Tensor vals = d["vals"].cast<Tensor>().contiguous();
+  c.vals.assign(vals.data_ptr<V>(), vals.data_ptr<V>() + vals.numel());
+  c.nnz = (sp::idx_t)c.vals.size();
+  for (int l = 0; l < c.nmodes; ++l)
+    c.nfibs[l] = l < c.nmodes - 1 ? (c.fptr[l].empty() ? 0 : c.fptr[l].size() - 1)
+                                  : c.nnz;
+  return c;
+}
+
+#define DTYPE_DISPATCH(vals_dtype, fn)                                     \
+  ((vals_dtype) == torch::kFloat64 ? fn(double) : fn(float))
+
+// ---------------------------------------------------------------- io
+
+static py::object py_tensor_load(const std::string & path, const std::string & dtype) {
+  if (dtype == "f32") {
+    auto tt = sp::tensor_load<float>(path);
+    auto [i, v, d] = coo_to_torch(tt);
+    return py::make_tuple(i, v, d);
+  }
+  auto tt = sp::tensor_load<double>(path);
+  auto [i, v, d] = coo_to_torch(tt);
+  return py::make_tuple(i, v, d);
+}
+
+static void py_tns_write(const std::string & path, Tensor inds, Tensor vals,
+                         std::vector<int64_t> dims) {
+  if (vals.scalar_type() == torch::kFloat32) {
+    auto tt = coo_from_torch<float>(inds, vals, dims);
+    sp::tns_write(tt, path);
+  } else {
+    auto tt = coo_from_torch<double>(inds, vals, dims);
+    sp::tns_write(tt, path);
+  }
+}
+
+static void py_bin_write(const std::string & path, Tensor inds, Tensor vals,
+                         std::vector<int64_t> dims, int idx_bytes, int val_bytes) {
+  if (vals.scalar_type() == torch::kFloat32) {
+    auto tt = coo_from_torch<float>(inds, vals, dims);
+    sp::bin_write(tt, path, idx_bytes, val_bytes);
+  } else {
+    auto tt = coo_from_torch<double>(inds, vals, dims);
+    sp::bin_write(tt, path, idx_bytes, val_bytes);
+  }
+}
+
+// ------------------------------------------------------------- coo utils
+
+template <typename V>
+static py::tuple py_coo_fix_t(Tensor inds, Tensor vals, std::vector<int64_t> dims,
+                              bool dedup, bool compress) {
+  auto tt = coo_from_torch<V>(inds, vals, dims);
+  int64_t ndups = 0, nempty = 0;
+  if (dedup) {
+    std::vector<int> perm(tt.nmodes);
+    for (int m = 0; m < tt.nmodes; ++m) perm[m] = m;
+    sp::coo_sort(tt, perm.data());
+    ndups = (int64_t)sp::coo_remove_dups(tt);
+  }
+  py::list indmaps;
+  if (compress) {
+    nempty = (int64_t)sp::coo_remove_empty(tt);
+    for (int m = 0; m < tt.nmodes; ++m) {
+      if (tt.indmap[m].empty()) {
+        indmaps.append(py::none());
+      } else {
+        Tensor t = torch::empty({(int64_t)tt.indmap[m].size()}, torch::kInt64);
+        std::copy(tt.indmap[m].begin(), tt.indmap[m].end(), t.data_ptr<int64_t>());
+        indmaps.append(t);
+      }
+    }
+  }
+  auto [i, v, d] = coo_to_torch(tt);
+  return py::make_tuple(i, v, d, ndups, nempty, indmaps);
+}
+
+static py::tuple py_coo_fix(Tensor inds, Tensor vals, std::vector<int64_t> dims,
+                            bool dedup, bool compress) {
+  if (vals.scalar_type() == torch::kFloat32)
+    return py_coo_fix_t<float>(inds, vals, dims, dedup, compress);
+  return py_coo_fix_t<double>(inds, vals, dims, dedup, compress);
+}
+
+// ------------------------------------------------------------- csf build
+
+template <typename V>
+static py::object py_csf_build_t(Tensor inds, Tensor vals,
+                                 std::vector<int64_t> dims,
+                                 std::vector<int64_t> perm) {
+  auto tt = coo_from_torch<V>(inds, vals, dims);
+  int p[sp::MAX_NMODES];
+  for (size_t i = 0; i < perm.size(); ++i) p[i] = (int)perm[i];
+  auto c = sp::csf_build(tt, p);
+  return csf_to_py(c);
+}
+
+static py::object py_csf_build(Tensor inds, Tensor vals,
+                               std::vector<int64_t> dims,
+                               std::vector<int64_t> perm) {
+  if (vals.scalar_type() == torch::kFloat32)
+    return py_csf_build_t<float>(inds, vals, dims, perm);
+  return py_csf_build_t<double>(inds, vals, dims, perm);
+}
+
+// ---------------------------------------------------------------- mttkrp
+
+template <typename V>
+static Tensor py_mttkrp_stream_t(Tensor inds, Tensor vals,
+                                 std::vector<int64_t> dims,
+                                 std::vector<Tensor> mats, int mode) {
+  auto tt = coo_from_torch<V>(inds, vals, dims);
+  const int F = (int)mats[0].size(1);
+  std::vector<const V*> mp;
+  std::vector<Tensor> mc;
+  for (auto & m : mats) { mc.push_back(m.contiguous()); mp.push_back(mc.back().data_ptr<V>()); }
+  Tensor out = torch::empty({dims[mode], F}, mats[0].options());
+  sp::mttkrp_stream(tt, mp.data(), out.data_ptr<V>(), mode, F);
+  return out;
+}
+
+static Tensor py_mttkrp_stream(Tensor inds, Tensor vals, std::vector<int64_t> dims,
+                               std::vector<Tensor> mats, int mode) {
+  if (vals.scalar_type() == torch::kFloat32)
+    return py_mttkrp_stream_t<float>(inds, vals, dims, mats, mode);
+  return py_mttkrp_stream_t<double>(inds, vals, dims, mats, mode);
+}
+
+template <typename V>
+static Tensor py_mttkrp_csf_t(py::dict csf, std::vector<Tensor> mats, int mode,
+                              int nthreads) {
+  auto c = csf_from_py<V>(csf);
+  const int F = (int)mats[0].size(1);
+  std::vector<const V*> mp;
+  std::vector<Tensor> mc;
+  for (auto & m : mats) { mc.push_back(m.contiguous()); mp.push_back(mc.back().data_ptr<V>()); }
+  Tensor out = torch::empty({(int64_t)c.dims[mode], F}, mats[0].options());
+  sp::mttkrp_csf_cpu(c, mp.data(), out.data_ptr<V>(), mode, F, nthreads);
+  return out;
+}
+
+static Tensor py_mttkrp_csf(py::dict csf, std::vector<Tensor> mats, int mode,
+                            int nthreads) {
+  if (mats[0].scalar_type() == torch::kFloat32)
+    return py_mttkrp_csf_t<float>(csf, mats, mode, nthreads);
+  return py_mttkrp_csf_t<double>(csf, mats, mode, nthreads);
+}
+
+// ------------------------------------------------------------------- cpd
+
+template <typename V>
+static py::dict py_cpd_als_t(Tensor inds, Tensor vals, std::vector<int64_t> dims,
+                             int rank, py::dict opts) {
+  auto o = opts_from_dict(opts);
+  auto tt = coo_from_torch<V>(inds, vals, dims);
+  auto set = sp::csf_alloc(tt, o);
+  auto k = sp::cpd_als(set, rank, o);
+  py::dict r;
+  py::list factors;
+  const auto vopt = torch::TensorOptions().dtype(
+      std::is_same<V, double>::value ? torch::kFloat64 : torch::kFloat32);
+  for (int m = 0; m < k.nmodes; ++m) {
+    Tensor t = torch::empty({(int64_t)k.dims[m], rank}, vopt);
+    std::copy(k.factors[m].begin(), k.factors[m].end(), t.data_ptr<V>());
+    factors.append(t);
+  }
+  Tensor lam = torch::empty({rank}, vopt);
+  std::copy(k.lambda.begin(), k.lambda.end(), lam.data_ptr<V>());
+  r["factors"] = factors;
+  r["lambda"] = lam;
+  r["fit"] = k.fit;
+  r["niters"] = k.niters;
+  return r;
+}
+
+static py::dict py_cpd_als(Tensor inds, Tensor vals, std::vector<int64_t> dims,
+                           int rank, py::dict opts) {
+  if (vals.scalar_type() == torch::kFloat32)
+    return py_cpd_als_t<float>(inds, vals, dims, rank, opts);
+  return py_cpd_als_t<double>(inds, vals, dims, rank, opts);
+}
+
+// --------------------------------------------------------------- dense ops
+
+static Tensor py_seeded_init(int64_t nrows, int rank, int64_t row0,
+                             uint64_t seed, int mode, const std::string & dtype) {
+  if (dtype == "f32") {
+    Tensor t = torch::empty({nrows, rank}, torch::kFloat32);
+    sp::seeded_factor_init(t.data_ptr<float>(), (sp::idx_t)nrows, rank,
+                           (sp::idx_t)row0, seed, mode);
+    return t;
+  }
+  Tensor t = torch::empty({nrows, rank}, torch::kFloat64);
+  sp::seeded_factor_init(t.data_ptr<double>(), (sp::idx_t)nrows, rank,
+                         (sp::idx_t)row0, seed, mode);
+  return t;
+}
+
+static std::vector<int64_t> py_order_modes(std::vector<int64_t> dims,
+                                           const std::string & policy, int mode) {
+  const int nm = (int)dims.size();
+  sp::idx_t d[sp::MAX_NMODES];
+  for (int m = 0; m < nm; ++m) d[m] = (sp::idx_t)dims[m];
+  int p[sp::MAX_NMODES];
+  if (policy == "smallfirst") sp::order_smallfirst(d, nm, p);
+  else if (policy == "root") sp::order_root(d, nm, mode, p);
+  else sp::order_leaf(d, nm, mode, p);
+  return std::vector<int64_t>(p, p + nm);
+}
+
+// --------------------------------------------------- HIP launcher externs
+
+extern "C" {
+// defined in csrc/hip/mttkrp_kernels.hip (gfx950); stream is hipStream_t
+void splatt_hip_mttkrp_root3_f64(
+    const int64_t* fptr0, const int32_t* fids0, const int64_t* fptr1,
+    const int32_t* fids1, const int32_t* fids2, const double* vals,
+    int64_t nslices, int64_t nfibs, int64_t nnz,
+    const double* A1, const double* A2, double* out, int rank, void* stream);
+void splatt_hip_mttkrp_root3_f32(
+    const int64_t* fptr0, const int32_t* fids0, const int64_t* fptr1,
+    const int32_t* fids1, const int32_t* fids2, const float* vals,
+    int64_t nslices, int64_t nfibs, int64_t nnz,
+    const float* A1, const float* A2, float* out, int rank, void* stream);
+void splatt_hip_mttkrp_intl3_f64(
+    const int64_t* fptr0, const int32_t* fids0, const int64_t* fptr1,
+    const int32_t* fids1, const int32_t* fids2, const double* vals,
+    int64_t nslices, int64_t nfibs, int64_t nnz,
+    const double* A0, const double* A2, double* out, int rank, void* stream);
+void splatt_hip_mttkrp_intl3_f32(
+    const int64_t* fptr0, const int32_t* fids0, const int64_t* fptr1,
+    const int32_t* fids1, const int32_t* fids2, const float* vals,
+    int64_t nslices, int64_t nfibs, int64_t nnz,
+    const float* A0, const float* A2, float* out, int rank, void* stream);
+void splatt_hip_mttkrp_leaf3_f64(
+    const int64_t* fptr0, const int32_t* fids0, const int64_t* fptr1,
+    const int32_t* fids1, const int32_t* fids2, const double* vals,
+    int64_t nslices, int64_t nfibs, int64_t nnz,
+    const double* A0, const double* A1, double* out, int rank, void* stream);
+void splatt_hip_mttkrp_leaf3_f32(
+    const int64_t* fptr0, const int32_t* fids0, const int64_t* fptr1,
+    const int32_t* fids1, const int32_t* fids2, const float* vals,
+    int64_t nslices, int64_t nfibs, int64_t nnz,
+    const float* A0, const float* A1, float* out, int rank, void* stream);
+int splatt_hip_kernels_arch(void);
+}
+
+// thin wrappers: Python passes contiguous CUDA tensors + stream handle
+template <typename V>
+static void gpu_mttkrp3_t(int which, Tensor fptr0, py::object fids0, Tensor fptr1,
+                          Tensor fids1, Tensor fids2, Tensor vals,
+                          Tensor Ma, Tensor Mb, Tensor out, int64_t stream) {
+  const int64_t nslices = fptr0.numel() - 1;
+  const int64_t nfibs = fptr1.numel() - 1;
+  const int64_t nnz = vals.numel();
+  const int rank = (int)Ma.size(1);
+  const int32_t * f0 = fids0.is_none() ? nullptr
+                       : fids0.cast<Tensor>().data_ptr<int32_t>();
+  const bool f64 = std::is_same<V, double>::value;
+  auto fp0 = fptr0.data_ptr<int64_t>();
+  auto fp1 = fptr1.data_ptr<int64_t>();
+  auto fi1 = fids1.data_ptr<int32_t>();
+  auto fi2 = fids2.data_ptr<int32_t>();
+  auto vp = vals.data_ptr<V>();
+  auto a = Ma.data_ptr<V>();
+  auto b = Mb.data_ptr<V>();
+  auto o = out.data_ptr<V>();
+  void * s = (void*)stream;
+  if (f64) {
+    auto vd = (const double*)vp; auto ad = (const double*)a;
+    auto bd = (const double*)b; auto od = (double*)o;
+    if (which == 0) splatt_hip_mttkrp_root3_f64(fp0, f0, fp1, fi1, fi2, vd, nslices, nfibs, nnz, ad, bd, od, rank, s);
+    else if (which == 1) splatt_hip_mttkrp_intl3_f64(fp0, f0, fp1, fi1, fi2, vd, nslices, nfibs, nnz, ad, bd, od, rank, s);
+    else splatt_hip_mttkrp_leaf3_f64(fp0, f0, fp1, fi1, fi2, vd, nslices, nfibs, nnz, ad, bd, od, rank, s);
+  } else {
+    auto vf = (const float*)vp; auto af = (const float*)a;
+    auto bf = (const float*)b; auto of = (float*)o;
+    if (which == 0) splatt_hip_mttkrp_root3_f32(fp0, f0, fp1, fi1, fi2, vf, nslices, nfibs, nnz, af, bf, of, rank, s);
+    else if (which == 1) splatt_hip_mttkrp_intl3_f32(fp0, f0, fp1, fi1, fi2, vf, nslices, nfibs, nnz, af, bf, of, rank, s);
+    else splatt_hip_mttkrp_leaf3_f32(fp0, f0, fp1, fi1, fi2, vf, nslices, nfibs, nnz, af, bf, of, rank, s);
+  }
+}
+
+static void py_gpu_mttkrp3(int which, Tensor fptr0, py::object fids0, Tensor fptr1,
+                           Tensor fids1, Tensor fids2, Tensor vals,
+                           Tensor Ma, Tensor Mb, Tensor out, int64_t stream) {
+  if (vals.scalar_type() == torch::kFloat32)
+    gpu_mttkrp3_t<float>(which, fptr0, fids0, fptr1, fids1, fids2, vals, Ma, Mb, out, stream);
+  else
+    gpu_mttkrp3_t<double>(which, fptr0, fids0, fptr1, fids1, fids2, vals, Ma, Mb, out, stream);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("tensor_load", &py_tensor_load, "load .tns/.bin tensor");
+  m.def("tns_write", &py_tns_write);
+  m.def("bin_write", &py_bin_write);
+  m.def("coo_fix", &py_coo_fix, "sort+dedup and/or remove empty slices");
+  m.def("csf_build", &py_csf_build, "CPU CSF build for a level permutation");
+  m.def("mttkrp_stream", &py_mttkrp_stream, "COO gold-oracle MTTKRP (CPU)");
+  m.def("mttkrp_csf_cpu", &py_mttkrp_csf, "CSF MTTKRP (CPU)");
+  m.def("cpd_als_cpu", &py_cpd_als, "CPD-ALS on the CPU reference path");
+  m.def("seeded_init", &py_seeded_init, "partition-invariant seeded factor init");
+  m.def("order_modes", &py_order_modes, "CSF mode-order policies");
+  m.def("gpu_mttkrp3", &py_gpu_mttkrp3, "3-mode CSF MTTKRP HIP kernels");
+  m.def("hip_arch", []() { return splatt_hip_kernels_arch(); });
+}

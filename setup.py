@@ -1,0 +1,65 @@
+"""In-tree build for the MI355X-native sparse tensor engine.
+
+Two-stage build, no hipify, no dual paths:
+  1. hipcc --offload-arch=gfx950 compiles csrc/hip/*.hip to objects
+     (pure CDNA4 HIP; cross-compiles fine on a GPU-less box).
+  2. a torch CppExtension compiles the C++17 core + pybind boundary and
+     links the HIP objects against libamdhip64.
+
+Usage: python setup.py build_ext --inplace
+"""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CppExtension
+
+ROOT = Path(__file__).resolve().parent
+ROCM = Path(os.environ.get("ROCM_PATH", "/opt/rocm"))
+HIPCC = str(ROCM / "bin" / "hipcc")
+ARCH = os.environ.get("SPLATT_GPU_ARCH", "gfx950")
+
+HIP_SOURCES = sorted((ROOT / "csrc" / "hip").glob("*.hip"))
+CORE_SOURCES = sorted((ROOT / "csrc" / "core").glob("*.cpp"))
+
+
+def build_hip_objects():
+    objdir = ROOT / "build" / "hip_obj"
+    objdir.mkdir(parents=True, exist_ok=True)
+    objs = []
+    for src in HIP_SOURCES:
+        obj = objdir / (src.stem + ".o")
+        if not obj.exists() or obj.stat().st_mtime < src.stat().st_mtime:
+            cmd = [
+                HIPCC, f"--offload-arch={ARCH}", "-O3", "-std=c++17",
+                "-fPIC", "-munsafe-fp-atomics", "-c", str(src), "-o", str(obj),
+            ]
+            print("[hipcc]", " ".join(cmd), flush=True)
+            subprocess.check_call(cmd)
+        objs.append(str(obj))
+    return objs
+
+
+hip_objs = build_hip_objects()
+
+ext = CppExtension(
+    name="splatt_amd._C",
+    sources=[str(p.relative_to(ROOT)) for p in CORE_SOURCES]
+    + ["csrc/pybind.cpp"],
+    include_dirs=[str(ROOT / "csrc")],
+    extra_compile_args=["-O3", "-std=c++17", "-fopenmp", "-march=native"],
+    extra_objects=hip_objs,
+    extra_link_args=["-fopenmp", f"-L{ROCM}/lib", "-lamdhip64",
+                     f"-Wl,-rpath,{ROCM}/lib"],
+)
+
+setup(
+    name="splatt_amd",
+    version="0.1.0",
+    packages=["splatt_amd", "splatt_amd.ops", "splatt_amd.parallel",
+              "splatt_amd.utils"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=False)},
+)

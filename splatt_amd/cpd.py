@@ -1,0 +1,140 @@
+"""CPD-ALS drivers.
+
+Capability parity: reference src/cpd.c (cpd_als_iterate:271-387, fit
+math:116-265, normalize schedule cpd.c:343-347, post-process:391-411).
+
+Two drivers:
+  * `cpd_als` — the device-resident driver: HIP MTTKRP + rocBLAS (via torch)
+    for Gram/solve/normalize. Factors stay in HBM for the whole run; only
+    lambda and the fit scalar cross to host. Also runs on CPU tensors (torch
+    CPU + C++ CSF MTTKRP), which is what the gloo multi-process tests use.
+  * `cpd_als_cpu_native` — the pure C++ reference path (`splatt cpd`
+    1-thread CPU config of BASELINE.json).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+from splatt_amd._ext import native
+from splatt_amd.csf import CsfSet, csf_alloc
+from splatt_amd.mttkrp import mttkrp
+from splatt_amd.sptensor import SpTensor
+
+
+@dataclass
+class CpdOptions:
+    tolerance: float = 1e-5
+    max_iters: int = 50
+    seed: int = 0x5EED5EED
+    csf_alloc: str = "two"
+    nthreads: int = 0
+    verbose: bool = False
+
+
+@dataclass
+class Kruskal:
+    factors: List[torch.Tensor]
+    lam: torch.Tensor
+    fit: float
+    niters: int
+    fit_trace: List[float] = field(default_factory=list)
+
+
+def seeded_init(nrows: int, rank: int, mode: int, seed: int,
+                row0: int = 0, dtype: torch.dtype = torch.float64) -> torch.Tensor:
+    """Partition-invariant factor init: element (i,f) depends only on
+    (seed, mode, global row i, f) — the property the reference gets from
+    root-generated mpi_mat_rand (mpi/mpi_io.c:1097-1176)."""
+    tag = "f32" if dtype == torch.float32 else "f64"
+    return native().seeded_init(nrows, rank, row0, seed, mode, tag)
+
+
+def _normalize(A: torch.Tensor, it: int) -> torch.Tensor:
+    if it == 0:
+        lam = A.square().sum(dim=0).sqrt()
+    else:
+        lam = A.abs().amax(dim=0).clamp_(min=1.0)
+    lam = torch.where(lam == 0, torch.ones_like(lam), lam)
+    A /= lam
+    return lam
+
+
+def cpd_als(src: CsfSet | SpTensor, rank: int,
+            opts: Optional[CpdOptions] = None) -> Kruskal:
+    opts = opts or CpdOptions()
+    cs = src if isinstance(src, CsfSet) else csf_alloc(src, opts.csf_alloc)
+    nm = cs.nmodes
+    dims = cs.dims
+    dev = cs.csfs[0].device
+    dtype = cs.csfs[0].vals.dtype
+
+    factors = [seeded_init(dims[m], rank, m, opts.seed, dtype=dtype).to(dev)
+               for m in range(nm)]
+    grams = [f.T @ f for f in factors]
+    norm_x = float(cs.csfs[0].vals.double().square().sum())
+    lam = torch.ones(rank, dtype=dtype, device=dev)
+    buf = torch.empty(max(dims), rank, dtype=dtype, device=dev)
+
+    fit = old_fit = 0.0
+    trace: List[float] = []
+    niters = 0
+    ones = torch.ones(rank, rank, dtype=dtype, device=dev)
+
+    for it in range(opts.max_iters):
+        for m in range(nm):
+            mb = buf[: dims[m]]
+            mttkrp(cs, factors, m, out=mb, nthreads=opts.nthreads)
+            G = ones.clone()
+            for o in range(nm):
+                if o != m:
+                    G *= grams[o]
+            # solve A * G = mttkrp  =>  G^T A^T = mttkrp^T (G symmetric)
+            L = torch.linalg.cholesky(
+                G + 1e-12 * torch.eye(rank, dtype=dtype, device=dev) * G.diagonal().abs().max())
+            A = torch.cholesky_solve(mb.T, L).T.contiguous()
+            lam = _normalize(A, it)
+            factors[m] = A
+            grams[m] = A.T @ A
+
+        # fit from last mode's pre-solve MTTKRP output (reference trick):
+        # <X,K> = sum_f lam_f * sum_i buf[i,f] * A_last[i,f]
+        mlast = nm - 1
+        inner = float((buf[: dims[mlast]].double()
+                       * factors[mlast].double()).sum(dim=0) @ lam.double())
+        Gall = ones.clone()
+        for o in range(nm):
+            Gall *= grams[o]
+        knorm = float(lam.double() @ Gall.double() @ lam.double())
+        residual = math.sqrt(max(0.0, norm_x + knorm - 2 * inner))
+        fit = 1.0 - residual / math.sqrt(norm_x)
+        trace.append(fit)
+        niters = it + 1
+        if opts.verbose:
+            print(f"  its = {it + 1} fit = {fit:.5f} delta = {fit - old_fit:+.4e}")
+        if it > 0 and abs(fit - old_fit) < opts.tolerance:
+            break
+        old_fit = fit
+
+    return Kruskal(factors=factors, lam=lam, fit=fit, niters=niters,
+                   fit_trace=trace)
+
+
+def cpd_als_cpu_native(t: SpTensor, rank: int,
+                       opts: Optional[CpdOptions] = None) -> Kruskal:
+    """The C++ host reference path (BASELINE config 1: 1-thread CPU cpd)."""
+    opts = opts or CpdOptions()
+    r = native().cpd_als_cpu(
+        t.inds.cpu(), t.vals.cpu(), list(t.dims), rank,
+        {
+            "tolerance": opts.tolerance,
+            "max_iters": opts.max_iters,
+            "seed": opts.seed,
+            "csf_alloc": opts.csf_alloc,
+            "nthreads": opts.nthreads,
+        })
+    return Kruskal(factors=list(r["factors"]), lam=r["lambda"],
+                   fit=float(r["fit"]), niters=int(r["niters"]))

@@ -1,0 +1,64 @@
+"""CSF structural invariants (reference tests/csf_test.c:34-199):
+nnz conservation, leaf fids == sorted COO order, dense-root label dropping,
+mode-order policies, frobsq, and CPU-vs-device-algorithm builder equality."""
+import torch
+
+import splatt_amd as sp
+from splatt_amd.csf import _build_csf_device, build_csf
+
+
+def test_mode_orders():
+    dims = [100, 10, 50]
+    assert sp.order_modes(dims, "smallfirst") == [1, 2, 0]
+    assert sp.order_modes(dims, "root", 0) == [0, 1, 2]
+    assert sp.order_modes(dims, "leaf", 1) == [2, 0, 1]
+
+
+def test_structure_invariants(small3):
+    cs = sp.csf_alloc(small3, "two")
+    for c in cs.csfs:
+        nm = c.nmodes
+        assert c.nnz == small3.nnz
+        assert int(c.fptr[0][-1]) == c.nfibs(1)
+        for l in range(nm - 1):
+            fp = c.fptr[l]
+            assert bool((fp[1:] > fp[:-1]).all()), "no empty nodes"
+        # leaf fids are the sorted COO leaf column
+        leafmode = c.dim_perm[-1]
+        svals = c.vals
+        assert svals.numel() == small3.nnz
+        # per-mode nnz histogram is preserved
+        hist_ref = torch.bincount(small3.inds[leafmode],
+                                  minlength=small3.dims[leafmode])
+        hist_csf = torch.bincount(c.fids[nm - 1].long(),
+                                  minlength=small3.dims[leafmode])
+        assert torch.equal(hist_ref, hist_csf)
+        # values conserved as a multiset (sum & sumsq)
+        assert abs(float(svals.sum() - small3.vals.sum())) < 1e-9
+
+
+def test_dense_root_labels_dropped():
+    t = sp.SpTensor.synthetic([4, 300, 300], 20000, seed=3)
+    c = build_csf(t, [0, 1, 2])
+    assert c.fids[0] is None  # all 4 slices populated => identity root
+
+
+def test_frobsq(small3):
+    cs = sp.csf_alloc(small3, "one")
+    got = float(cs.csfs[0].vals.double().square().sum())
+    want = small3.normsq()
+    assert abs(got - want) < 1e-9
+
+
+def test_device_algorithm_matches_cpu(small3, med4):
+    for t in (small3, med4):
+        for policy in ("smallfirst",):
+            perm = sp.order_modes(t.dims, policy)
+            a = build_csf(t, perm)
+            b = _build_csf_device(t, perm)
+            for l in range(t.nmodes):
+                for x, y in ((a.fptr[l], b.fptr[l]), (a.fids[l], b.fids[l])):
+                    assert (x is None) == (y is None)
+                    if x is not None:
+                        assert torch.equal(x, y)
+            assert torch.equal(a.vals, b.vals)

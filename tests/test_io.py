@@ -1,0 +1,62 @@
+"""I/O tests (reference tests/io_test.c): text round-trip, 0/1-index
+equivalence, binary round-trip incl. width down-conversion."""
+import torch
+
+import splatt_amd as sp
+
+
+def test_tns_roundtrip(tmp_path, small3):
+    p = tmp_path / "t.tns"
+    small3.save(p)
+    t2 = sp.load(p)
+    assert t2.dims == small3.dims
+    assert torch.equal(t2.inds, small3.inds)
+    assert (t2.vals - small3.vals).abs().max() < 1e-15
+
+
+def test_zero_vs_one_indexed(tmp_path):
+    t = sp.SpTensor.synthetic([10, 12, 9], 200, seed=5)
+    # ensure index 0 occurs so autodetect sees a 0-based file
+    t.inds[0, 0] = 0
+    one = tmp_path / "one.tns"
+    zero = tmp_path / "zero.tns"
+    with open(one, "w") as f:
+        for i in range(t.nnz):
+            f.write(" ".join(str(int(t.inds[m, i]) + 1) for m in range(3))
+                    + f" {float(t.vals[i]):.17g}\n")
+    with open(zero, "w") as f:
+        for i in range(t.nnz):
+            f.write(" ".join(str(int(t.inds[m, i])) for m in range(3))
+                    + f" {float(t.vals[i]):.17g}\n")
+    a, b = sp.load(one), sp.load(zero)
+    assert torch.equal(a.inds, b.inds)
+    assert a.dims == b.dims
+
+
+def test_bin_roundtrip(tmp_path, small3):
+    p = tmp_path / "t.bin"
+    small3.save(p)
+    t2 = sp.load(p)
+    assert t2.dims == small3.dims
+    assert torch.equal(t2.inds, small3.inds)
+    assert torch.equal(t2.vals, small3.vals)
+
+
+def test_bin_width_downconvert(tmp_path, small3):
+    from splatt_amd._ext import native
+    p = str(tmp_path / "t32.bin")
+    native().bin_write(p, small3.inds, small3.vals, list(small3.dims), 4, 4)
+    t2 = sp.load(p)
+    assert torch.equal(t2.inds, small3.inds)
+    assert (t2.vals - small3.vals).abs().max() < 1e-6
+
+
+def test_fixed_dedup_compress():
+    inds = torch.tensor([[1, 1, 5, 5], [2, 2, 3, 3], [0, 0, 7, 7]])
+    vals = torch.tensor([1.0, 2.0, 3.0, 4.0], dtype=torch.float64)
+    t = sp.SpTensor(inds, vals, [10, 10, 10])
+    f = t.fixed(dedup=True, compress=True)
+    assert f.nnz == 2
+    assert sorted(f.vals.tolist()) == [3.0, 7.0]
+    assert f.dims == [2, 2, 2]  # empty slices removed
+    assert f.indmaps[0].tolist() == [1, 5]
