@@ -1,0 +1,80 @@
+"""Multi-process distributed CPD over gloo (CPU, world_size 2) — validates
+the collective schedule that runs over RCCL on the GPU node. The key
+property (reference design: rank-invariant mpi_mat_rand, mpi/mpi_io.c:1097):
+fit at world N == fit at world 1 for the same seed."""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+import splatt_amd as sp
+from splatt_amd.parallel.dist_cpd import (
+    build_shard_csf, dist_cpd_als, localize_shard, partition_rows)
+
+DIMS = [40, 30, 60]
+NNZ = 6000
+RANK_F = 8
+ITERS = 6
+SEED = 77
+
+
+def test_partition_rows():
+    assert partition_rows(10, 4, 0) == (0, 3)
+    assert partition_rows(10, 4, 1) == (3, 3)
+    assert partition_rows(10, 4, 2) == (6, 2)
+    assert partition_rows(10, 4, 3) == (8, 2)
+    total = sum(partition_rows(97, 7, r)[1] for r in range(7))
+    assert total == 97
+
+
+def test_localize_shard():
+    t = sp.SpTensor.synthetic(DIMS, 500, seed=3)
+    row0, nloc = partition_rows(DIMS[2], 2, 1)
+    s = localize_shard(t, 2, row0, nloc)
+    assert s.dims[2] == nloc
+    assert int(s.inds[2].max()) < nloc
+    n0 = localize_shard(t, 2, 0, partition_rows(DIMS[2], 2, 0)[1]).nnz
+    assert n0 + s.nnz == t.nnz
+
+
+def _worker(rank, world, file_store, result_q):
+    torch.distributed.init_process_group(
+        "gloo", init_method=f"file://{file_store}", rank=rank,
+        world_size=world)
+    try:
+        t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+        part_mode = 2
+        row0, nloc = partition_rows(DIMS[part_mode], world, rank)
+        shard = localize_shard(t, part_mode, row0, nloc)
+        cs = build_shard_csf(shard, list(DIMS), "two")
+        opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+        k = dist_cpd_als(cs, part_mode, row0, list(DIMS), RANK_F, opts)
+        if rank == 0:
+            result_q.put(("fit", k.fit, k.niters))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dist_fit_matches_single_process(tmp_path):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    # single-process reference
+    t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+    opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+    k1 = sp.cpd_als(t, RANK_F, opts)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    store = str(tmp_path / "store")
+    procs = [ctx.Process(target=_worker, args=(r, 2, store, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    tag, fit2, niters2 = q.get()
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert tag == "fit"
+    assert niters2 == k1.niters
+    assert abs(fit2 - k1.fit) < 1e-8, (fit2, k1.fit)
