@@ -383,6 +383,45 @@ void splatt_hip_mttkrp_leaf3_f32(
     int64_t nslices, int64_t nfibs, int64_t nnz,
     const float* A0, const float* A1, float* out, int rank, void* stream);
 int splatt_hip_kernels_arch(void);
+// flat (expanded-CSF) kernels, csrc/hip/mttkrp_flat.hip
+void splatt_hip_mttkrp_flat_f64(
+    const int32_t*, const int32_t*, const int32_t*, const int32_t*,
+    const int32_t*, const double*, const double*, const double*,
+    const double*, const double*, int64_t, double*, int, int, void*);
+void splatt_hip_mttkrp_flat_f32(
+    const int32_t*, const int32_t*, const int32_t*, const int32_t*,
+    const int32_t*, const float*, const float*, const float*,
+    const float*, const float*, int64_t, float*, int, int, void*);
+}
+
+// key + per-other-level (idx, mat) pairs; nnz products folded by key runs
+static void py_gpu_mttkrp_flat(Tensor key, std::vector<Tensor> idx,
+                               std::vector<Tensor> mats, Tensor vals,
+                               Tensor out, int64_t stream) {
+  const int nother = (int)idx.size();
+  TORCH_CHECK(nother >= 2 && nother <= 4, "flat kernel supports 3..5 modes");
+  TORCH_CHECK((int)mats.size() == nother);
+  const int rank = (int)mats[0].size(1);
+  const int64_t nnz = vals.numel();
+  const int32_t * ip[4] = {nullptr, nullptr, nullptr, nullptr};
+  for (int t = 0; t < nother; ++t) ip[t] = idx[t].data_ptr<int32_t>();
+  if (vals.scalar_type() == torch::kFloat64) {
+    const double * mp[4] = {nullptr, nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
+    splatt_hip_mttkrp_flat_f64(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
+                               ip[3], mp[0], mp[1], mp[2], mp[3],
+                               vals.data_ptr<double>(), nnz,
+                               out.data_ptr<double>(), rank, nother,
+                               (void*)stream);
+  } else {
+    const float * mp[4] = {nullptr, nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+    splatt_hip_mttkrp_flat_f32(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
+                               ip[3], mp[0], mp[1], mp[2], mp[3],
+                               vals.data_ptr<float>(), nnz,
+                               out.data_ptr<float>(), rank, nother,
+                               (void*)stream);
+  }
 }
 
 // thin wrappers: Python passes contiguous CUDA tensors + stream handle
@@ -442,5 +481,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("seeded_init", &py_seeded_init, "partition-invariant seeded factor init");
   m.def("order_modes", &py_order_modes, "CSF mode-order policies");
   m.def("gpu_mttkrp3", &py_gpu_mttkrp3, "3-mode CSF MTTKRP HIP kernels");
+  m.def("gpu_mttkrp_flat", &py_gpu_mttkrp_flat,
+        "flat expanded-CSF MTTKRP HIP kernel (3..5 modes)");
   m.def("hip_arch", []() { return splatt_hip_kernels_arch(); });
 }

@@ -71,6 +71,33 @@ class Csf:
                 b += t.numel() * t.element_size()
         return b
 
+    def ancestor_expand(self, level: int) -> torch.Tensor:
+        """Per-nonzero ancestor label at `level` (int32, length nnz) — the
+        sorted coordinate column recovered from the CSF compression. Feeds
+        the flat MTTKRP kernels; cached after first use."""
+        cache = getattr(self, "_expand_cache", None)
+        if cache is None:
+            cache = {}
+            object.__setattr__(self, "_expand_cache", cache)
+        if level in cache:
+            return cache[level]
+        nm = self.nmodes
+        if level == nm - 1:
+            t = self.fids[nm - 1]
+        else:
+            # nnz start of each node at `level`: compose fptr chains down
+            nnzstart = self.fptr[nm - 2]
+            for l in range(nm - 3, level - 1, -1):
+                nnzstart = nnzstart[self.fptr[l]]
+            counts = nnzstart[1:] - nnzstart[:-1]
+            labels = self.fids[level]
+            if labels is None:  # dense root: identity labels
+                labels = torch.arange(counts.numel(), dtype=torch.int32,
+                                      device=counts.device)
+            t = torch.repeat_interleave(labels, counts)
+        cache[level] = t
+        return t
+
     def to_dict(self) -> dict:
         return {
             "fptr": [t.cpu() if t is not None else None for t in self.fptr],

@@ -42,10 +42,31 @@ def _gpu_mttkrp_csf(c: Csf, depth: int, mats: List[torch.Tensor],
         c.vals, ma.contiguous(), mb.contiguous(), out, stream)
 
 
+def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
+                     out: torch.Tensor) -> None:
+    """Flat expanded-CSF kernel (see csrc/hip/mttkrp_flat.hip): per-nnz
+    product of the non-output modes' rows folded by runs of the output key."""
+    nm = c.nmodes
+    key = c.ancestor_expand(depth)
+    idx, ms = [], []
+    for l in range(nm):
+        if l == depth:
+            continue
+        idx.append(c.ancestor_expand(l))
+        ms.append(mats[c.dim_perm[l]].contiguous())
+    stream = torch.cuda.current_stream().cuda_stream
+    native().gpu_mttkrp_flat(key, idx, ms, c.vals, out, stream)
+
+
 def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
            out: Optional[torch.Tensor] = None,
-           nthreads: int = 0) -> torch.Tensor:
-    """MTTKRP for output `mode`; `mats` indexed by tensor mode."""
+           nthreads: int = 0, alg: str = "flat") -> torch.Tensor:
+    """MTTKRP for output `mode`; `mats` indexed by tensor mode.
+
+    Device algorithms: 'flat' (default, expanded-CSF streaming kernel) or
+    'csf' (hierarchical fiber-walk kernels, 3-mode only) — the reference
+    keeps multiple MTTKRP algorithms selectable the same way (bench.c).
+    """
     if isinstance(src, CsfSet):
         c = src.csfs[src.mode_csf[mode]]
         depth = src.mode_depth[mode]
@@ -61,7 +82,10 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
         if native().hip_arch() != 950:
             raise RuntimeError("HIP kernels not built for gfx950")
         out.zero_()
-        _gpu_mttkrp_csf(c, depth, mats, mode, out)
+        if alg == "flat":
+            _gpu_mttkrp_flat(c, depth, mats, out)
+        else:
+            _gpu_mttkrp_csf(c, depth, mats, mode, out)
         return out
     res = native().mttkrp_csf_cpu(c.to_dict(), [m.cpu() for m in mats],
                                   mode, nthreads)

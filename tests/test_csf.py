@@ -62,3 +62,20 @@ def test_device_algorithm_matches_cpu(small3, med4):
                     if x is not None:
                         assert torch.equal(x, y)
             assert torch.equal(a.vals, b.vals)
+
+
+def test_ancestor_expand_recovers_sorted_coords(small3, med4):
+    """expand[l][p] must equal the sorted COO coordinate of level l at p."""
+    from splatt_amd._ext import native
+    for t in (small3, med4):
+        perm = sp.order_modes(t.dims, "smallfirst")
+        c = build_csf(t, perm)
+        # reconstruct sorted coords independently: lexicographic argsort
+        keys = [t.inds[m] for m in perm]
+        order = torch.arange(t.nnz)
+        for lv in reversed(range(t.nmodes)):
+            order = order[torch.argsort(keys[lv][order], stable=True)]
+        for lv in range(t.nmodes):
+            exp = c.ancestor_expand(lv)
+            want = keys[lv][order].to(torch.int32)
+            assert torch.equal(exp, want), lv

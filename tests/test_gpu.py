@@ -26,17 +26,31 @@ def test_native_arch():
     assert native().hip_arch() == 950
 
 
+@pytest.mark.parametrize("alg", ["flat", "csf"])
 @pytest.mark.parametrize("policy", POLICIES)
 @pytest.mark.parametrize("rank", [16, 32, 64])
-def test_gpu_mttkrp_matches_oracle(t3, policy, rank):
+def test_gpu_mttkrp_matches_oracle(t3, policy, rank, alg):
     mats_c = make_mats(t3.dims, rank)
     mats_g = [m.cuda() for m in mats_c]
     cs = sp.csf_alloc(t3.to("cuda"), policy)
     for mode in range(3):
-        out = sp.mttkrp(cs, mats_g, mode)
+        out = sp.mttkrp(cs, mats_g, mode, alg=alg)
         ref = sp.mttkrp_stream(t3, mats_c, mode)
         err = (out.cpu() - ref).abs().max().item()
-        assert err < 1e-8, (policy, rank, mode, err)
+        assert err < 1e-8, (policy, rank, mode, alg, err)
+
+
+@pytest.mark.parametrize("nm_dims,nnz", [([60, 50, 70, 40], 80_000),
+                                         ([30, 25, 35, 20, 15], 60_000)])
+def test_gpu_mttkrp_flat_nmode(nm_dims, nnz):
+    t = sp.SpTensor.synthetic(nm_dims, nnz, seed=31)
+    mats_c = make_mats(t.dims, 16)
+    mats_g = [m.cuda() for m in mats_c]
+    cs = sp.csf_alloc(t.to("cuda"), "two")
+    for mode in range(len(nm_dims)):
+        out = sp.mttkrp(cs, mats_g, mode)
+        ref = sp.mttkrp_stream(t, mats_c, mode)
+        assert (out.cpu() - ref).abs().max().item() < 1e-8, mode
 
 
 def test_gpu_mttkrp_generic_rank(t3):
