@@ -70,7 +70,7 @@ def main():
     ap.add_argument("--rank-f", type=int, default=0, help="override CP rank")
     ap.add_argument("--dtype", default="f64", choices=["f64", "f32"])
     ap.add_argument("--device", default="cuda")
-    ap.add_argument("--csf", default="two", choices=["one", "two", "all"])
+    ap.add_argument("--csf", default="all", choices=["one", "two", "all"])
     args = ap.parse_args()
 
     dims, nnz_shard, rank_f, conc = CONFIGS[args.config]

@@ -21,6 +21,7 @@
 // {4,8,16,32,64}, generic otherwise).
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 namespace {
 
@@ -163,6 +164,13 @@ inline bool spec_ok(int F) {
   return F == 4 || F == 8 || F == 16 || F == 32 || F == 64;
 }
 
+inline int pick_unroll() {
+  // A/B lever for the memory-level-parallelism depth (default 8)
+  const char * e = getenv("SPLATT_MTTKRP_U");
+  const int u = e ? atoi(e) : 8;
+  return (u == 4 || u == 16) ? u : 8;
+}
+
 template <typename V>
 void launch_flat(const int32_t * key, const int32_t * const idx[4],
                  const V * const mats[4], const V * vals, int64_t nnz,
@@ -175,38 +183,27 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
 
 #define ARGS key, idx[0], idx[1], idx[2], idx[3], mats[0], mats[1], mats[2], \
              mats[3], vals, nnz, span, out
+#define L1(F_, N_, U_) \
+  hipLaunchKernelGGL((mttkrp_flat_kern<V, F_, N_, U_>), grid, block, 0, st, ARGS)
+#define LU(F_, N_) \
+  switch (uu) { case 4: L1(F_, N_, 4); break; case 16: L1(F_, N_, 16); break; \
+                default: L1(F_, N_, 8); break; }
+#define LF(N_) \
+  switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
+                  case 16: LU(16, N_); break; case 32: LU(32, N_); break; \
+                  default: LU(64, N_); break; }
   if (spec_ok(rank)) {
+    const int uu = pick_unroll();
     switch (nother) {
-      case 2:
-        switch (rank) {
-          case 4:  hipLaunchKernelGGL((mttkrp_flat_kern<V, 4, 2>),  grid, block, 0, st, ARGS); break;
-          case 8:  hipLaunchKernelGGL((mttkrp_flat_kern<V, 8, 2>),  grid, block, 0, st, ARGS); break;
-          case 16: hipLaunchKernelGGL((mttkrp_flat_kern<V, 16, 2>), grid, block, 0, st, ARGS); break;
-          case 32: hipLaunchKernelGGL((mttkrp_flat_kern<V, 32, 2>), grid, block, 0, st, ARGS); break;
-          default: hipLaunchKernelGGL((mttkrp_flat_kern<V, 64, 2>), grid, block, 0, st, ARGS); break;
-        }
-        break;
-      case 3:
-        switch (rank) {
-          case 4:  hipLaunchKernelGGL((mttkrp_flat_kern<V, 4, 3>),  grid, block, 0, st, ARGS); break;
-          case 8:  hipLaunchKernelGGL((mttkrp_flat_kern<V, 8, 3>),  grid, block, 0, st, ARGS); break;
-          case 16: hipLaunchKernelGGL((mttkrp_flat_kern<V, 16, 3>), grid, block, 0, st, ARGS); break;
-          case 32: hipLaunchKernelGGL((mttkrp_flat_kern<V, 32, 3>), grid, block, 0, st, ARGS); break;
-          default: hipLaunchKernelGGL((mttkrp_flat_kern<V, 64, 3>), grid, block, 0, st, ARGS); break;
-        }
-        break;
-      default:
-        switch (rank) {
-          case 4:  hipLaunchKernelGGL((mttkrp_flat_kern<V, 4, 4>),  grid, block, 0, st, ARGS); break;
-          case 8:  hipLaunchKernelGGL((mttkrp_flat_kern<V, 8, 4>),  grid, block, 0, st, ARGS); break;
-          case 16: hipLaunchKernelGGL((mttkrp_flat_kern<V, 16, 4>), grid, block, 0, st, ARGS); break;
-          case 32: hipLaunchKernelGGL((mttkrp_flat_kern<V, 32, 4>), grid, block, 0, st, ARGS); break;
-          default: hipLaunchKernelGGL((mttkrp_flat_kern<V, 64, 4>), grid, block, 0, st, ARGS); break;
-        }
-        break;
+      case 2: LF(2); break;
+      case 3: LF(3); break;
+      default: LF(4); break;
     }
     return;
   }
+#undef LF
+#undef LU
+#undef L1
 #undef ARGS
 #define GARGS key, idx[0], idx[1], idx[2], idx[3], mats[0], mats[1], mats[2], \
               mats[3], vals, nnz, span, rank, out

@@ -95,7 +95,9 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             # solve A * G = mttkrp  =>  G^T A^T = mttkrp^T (G symmetric)
             L = torch.linalg.cholesky(
                 G + 1e-12 * torch.eye(rank, dtype=dtype, device=dev) * G.diagonal().abs().max())
-            A = torch.cholesky_solve(mb.T, L).T.contiguous()
+            # F x F inverse once, then one well-shaped (n x F)(F x F) GEMM —
+            # beats a trsm against an n-row RHS at these tiny F
+            A = mb @ torch.cholesky_inverse(L)
             lam = _normalize(A, it)
             factors[m] = A
             grams[m] = A.T @ A

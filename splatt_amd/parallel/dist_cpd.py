@@ -168,7 +168,7 @@ def dist_cpd_step(st: DistCpdState, it: int) -> float:
             if o != m:
                 G *= st.grams[o]
         L = torch.linalg.cholesky(G + 1e-12 * G.diagonal().abs().max() * eye)
-        A = torch.cholesky_solve(mb.T, L).T.contiguous()
+        A = mb @ torch.cholesky_inverse(L)
         # normalize with GLOBAL column norms for the partitioned mode
         if m == q:
             if it == 0:
