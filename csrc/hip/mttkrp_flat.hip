@@ -111,6 +111,92 @@ mttkrp_flat_kern(const int32_t * __restrict__ key,
   atomic_add_g(&out[(int64_t)cur * F + c], acc);
 }
 
+// ------------------------------------------------- staged spec kernel (v2)
+// v1 has every lane of a column group redundantly load the same stream
+// words (key/idx/vals), so memory-level parallelism is capped by the VGPR
+// cost of the unroll. v2 stages streams COALESCED — lane slot c of a group
+// loads element pb+c of each stream in one instruction per F nonzeros —
+// and redistributes them with ds_bpermute (__shfl); the factor-row gathers
+// then issue in independent batches of 8, giving ~4x the outstanding
+// gathers per wave at lower VGPR pressure.
+template <typename V, int F, int NOTHER>
+__global__ void __launch_bounds__(256)
+mttkrp_flat2_kern(const int32_t * __restrict__ key,
+                  const int32_t * __restrict__ i0,
+                  const int32_t * __restrict__ i1,
+                  const int32_t * __restrict__ i2,
+                  const int32_t * __restrict__ i3,
+                  const V * __restrict__ m0, const V * __restrict__ m1,
+                  const V * __restrict__ m2, const V * __restrict__ m3,
+                  const V * __restrict__ vals, int64_t nnz, int64_t span,
+                  V * __restrict__ out) {
+  constexpr int R = WAVE / F;
+  constexpr int GB = (F >= 8) ? 8 : F;   // gather sub-batch
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + (threadIdx.x / WAVE);
+  const int c = lane % F;
+  const int g = lane / F;
+  const int gbase = g * F;
+  const int64_t w0 = wid * span;
+  if (w0 >= nnz) return;
+  const int64_t w1 = min64(nnz, w0 + span);
+  const int64_t gsz = (w1 - w0 + R - 1) / R;
+  const int64_t p0 = min64(w1, w0 + g * gsz);
+  const int64_t p1 = min64(w1, p0 + gsz);
+  if (p0 >= p1) return;
+
+  int32_t cur = key[p0];
+  V acc = (V)0;
+  for (int64_t pb = p0; pb < p1; pb += F) {
+    const int nb = (int)min64((int64_t)F, p1 - pb);       // group-uniform
+    const int64_t ps = pb + (c < nb ? c : nb - 1);        // clamped slot
+    const int32_t kreg = key[ps];
+    const int32_t i0reg = i0[ps];
+    const int32_t i1reg = i1[ps];
+    const int32_t i2reg = (NOTHER > 2) ? i2[ps] : 0;
+    const int32_t i3reg = (NOTHER > 3) ? i3[ps] : 0;
+    const V vreg = vals[ps];
+    for (int ub = 0; ub < nb; ub += GB) {
+      const int ne = nb - ub < GB ? nb - ub : GB;         // group-uniform
+      int32_t kk[GB];
+      V vv[GB], a0[GB], a1[GB], a2[GB], a3[GB];
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        const int src = gbase + (u < ne ? ub + u : ub);
+        kk[u] = __shfl(kreg, src, WAVE);
+        vv[u] = __shfl(vreg, src, WAVE);
+        const int32_t j0 = __shfl(i0reg, src, WAVE);
+        const int32_t j1 = __shfl(i1reg, src, WAVE);
+        a0[u] = m0[(int64_t)j0 * F + c];
+        a1[u] = m1[(int64_t)j1 * F + c];
+        if (NOTHER > 2) {
+          const int32_t j2 = __shfl(i2reg, src, WAVE);
+          a2[u] = m2[(int64_t)j2 * F + c];
+        }
+        if (NOTHER > 3) {
+          const int32_t j3 = __shfl(i3reg, src, WAVE);
+          a3[u] = m3[(int64_t)j3 * F + c];
+        }
+      }
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        if (u >= ne) break;
+        V x = vv[u] * a0[u] * a1[u];
+        if (NOTHER > 2) x *= a2[u];
+        if (NOTHER > 3) x *= a3[u];
+        if (kk[u] != cur) {
+          atomic_add_g(&out[(int64_t)cur * F + c], acc);
+          acc = (V)0;
+          cur = kk[u];
+        }
+        acc += x;
+      }
+    }
+  }
+  atomic_add_g(&out[(int64_t)cur * F + c], acc);
+}
+
 // ------------------------------------------------------ generic-rank kernel
 // lane = column (chunked by 64), wave walks its span serially. Correctness
 // path for ranks outside the spec set.
@@ -165,10 +251,10 @@ inline bool spec_ok(int F) {
 }
 
 inline int pick_unroll() {
-  // A/B lever for the memory-level-parallelism depth (default 8)
+  // A/B lever: 0 (default) = staged v2 kernel; 4/8/16 = v1 at that unroll
   const char * e = getenv("SPLATT_MTTKRP_U");
-  const int u = e ? atoi(e) : 8;
-  return (u == 4 || u == 16) ? u : 8;
+  const int u = e ? atoi(e) : 0;
+  return (u == 4 || u == 8 || u == 16) ? u : 0;
 }
 
 template <typename V>
@@ -185,9 +271,11 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
              mats[3], vals, nnz, span, out
 #define L1(F_, N_, U_) \
   hipLaunchKernelGGL((mttkrp_flat_kern<V, F_, N_, U_>), grid, block, 0, st, ARGS)
+#define L2K(F_, N_) \
+  hipLaunchKernelGGL((mttkrp_flat2_kern<V, F_, N_>), grid, block, 0, st, ARGS)
 #define LU(F_, N_) \
   switch (uu) { case 4: L1(F_, N_, 4); break; case 16: L1(F_, N_, 16); break; \
-                default: L1(F_, N_, 8); break; }
+                case 8: L1(F_, N_, 8); break; default: L2K(F_, N_); break; }
 #define LF(N_) \
   switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
                   case 16: LU(16, N_); break; case 32: LU(32, N_); break; \
