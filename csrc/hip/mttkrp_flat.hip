@@ -34,6 +34,13 @@ __device__ __forceinline__ void atomic_add_g(V * p, V v) {
   unsafeAtomicAdd(p, v);
 }
 
+// streamed-once arrays (key/idx/vals) are loaded non-temporally so they do
+// not evict the factor-row working set from the per-XCD L2s
+template <typename T>
+__device__ __forceinline__ T ldnt(const T * p) {
+  return __builtin_nontemporal_load(p);
+}
+
 // ---------------------------------------------------------- spec kernels
 // F lanes per column group, R = 64/F groups each walking a contiguous
 // sub-span; U-deep unroll for memory-level parallelism.
@@ -72,24 +79,24 @@ mttkrp_flat_kern(const int32_t * __restrict__ key,
     V prod[U];
     if (n == U) {
       #pragma unroll
-      for (int u = 0; u < U; ++u) k[u] = key[p + u];
+      for (int u = 0; u < U; ++u) k[u] = ldnt(&key[p + u]);
       #pragma unroll
       for (int u = 0; u < U; ++u) {
-        V x = vals[p + u] * m0[(int64_t)i0[p + u] * F + c]
-                          * m1[(int64_t)i1[p + u] * F + c];
-        if (NOTHER > 2) x *= m2[(int64_t)i2[p + u] * F + c];
-        if (NOTHER > 3) x *= m3[(int64_t)i3[p + u] * F + c];
+        V x = ldnt(&vals[p + u]) * m0[(int64_t)ldnt(&i0[p + u]) * F + c]
+                          * m1[(int64_t)ldnt(&i1[p + u]) * F + c];
+        if (NOTHER > 2) x *= m2[(int64_t)ldnt(&i2[p + u]) * F + c];
+        if (NOTHER > 3) x *= m3[(int64_t)ldnt(&i3[p + u]) * F + c];
         prod[u] = x;
       }
     } else {
       #pragma unroll
       for (int u = 0; u < U; ++u) {
         if (u < n) {
-          k[u] = key[p + u];
-          V x = vals[p + u] * m0[(int64_t)i0[p + u] * F + c]
-                            * m1[(int64_t)i1[p + u] * F + c];
-          if (NOTHER > 2) x *= m2[(int64_t)i2[p + u] * F + c];
-          if (NOTHER > 3) x *= m3[(int64_t)i3[p + u] * F + c];
+          k[u] = ldnt(&key[p + u]);
+          V x = ldnt(&vals[p + u]) * m0[(int64_t)ldnt(&i0[p + u]) * F + c]
+                            * m1[(int64_t)ldnt(&i1[p + u]) * F + c];
+          if (NOTHER > 2) x *= m2[(int64_t)ldnt(&i2[p + u]) * F + c];
+          if (NOTHER > 3) x *= m3[(int64_t)ldnt(&i3[p + u]) * F + c];
           prod[u] = x;
         } else {
           k[u] = cur;        // no-op in the fold
@@ -151,12 +158,12 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
   for (int64_t pb = p0; pb < p1; pb += F) {
     const int nb = (int)min64((int64_t)F, p1 - pb);       // group-uniform
     const int64_t ps = pb + (c < nb ? c : nb - 1);        // clamped slot
-    const int32_t kreg = key[ps];
-    const int32_t i0reg = i0[ps];
-    const int32_t i1reg = i1[ps];
-    const int32_t i2reg = (NOTHER > 2) ? i2[ps] : 0;
-    const int32_t i3reg = (NOTHER > 3) ? i3[ps] : 0;
-    const V vreg = vals[ps];
+    const int32_t kreg = ldnt(&key[ps]);
+    const int32_t i0reg = ldnt(&i0[ps]);
+    const int32_t i1reg = ldnt(&i1[ps]);
+    const int32_t i2reg = (NOTHER > 2) ? ldnt(&i2[ps]) : 0;
+    const int32_t i3reg = (NOTHER > 3) ? ldnt(&i3[ps]) : 0;
+    const V vreg = ldnt(&vals[ps]);
     for (int ub = 0; ub < nb; ub += GB) {
       const int ne = nb - ub < GB ? nb - ub : GB;         // group-uniform
       int32_t kk[GB];
