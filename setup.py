@@ -29,6 +29,7 @@ def build_hip_objects():
     objdir = ROOT / "build" / "hip_obj"
     objdir.mkdir(parents=True, exist_ok=True)
     objs = []
+    rebuilt = False
     for src in HIP_SOURCES:
         obj = objdir / (src.stem + ".o")
         if not obj.exists() or obj.stat().st_mtime < src.stat().st_mtime:
@@ -38,7 +39,14 @@ def build_hip_objects():
             ]
             print("[hipcc]", " ".join(cmd), flush=True)
             subprocess.check_call(cmd)
+            rebuilt = True
         objs.append(str(obj))
+    if rebuilt:
+        # setuptools does not track extra_objects mtimes: force a relink by
+        # dirtying the binding TU (else a stale _C.so keeps the old kernels)
+        (ROOT / "csrc" / "pybind.cpp").touch()
+        for so in (ROOT / "splatt_amd").glob("_C*.so"):
+            so.unlink()
     return objs
 
 
