@@ -392,6 +392,21 @@ void splatt_hip_mttkrp_flat_f32(
     const int32_t*, const int32_t*, const int32_t*, const int32_t*,
     const int32_t*, const float*, const float*, const float*,
     const float*, const float*, int64_t, float*, int, int, void*);
+// dense kernels, csrc/hip/dense_kernels.hip
+void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
+void splatt_hip_gram_f32(const float*, int64_t, int, float*, void*);
+}
+
+// G (FxF, pre-zeroed) += A^T A for tall-skinny device A
+static void py_gpu_gram(Tensor A, Tensor G, int64_t stream) {
+  const int64_t n = A.size(0);
+  const int F = (int)A.size(1);
+  if (A.scalar_type() == torch::kFloat64)
+    splatt_hip_gram_f64(A.data_ptr<double>(), n, F, G.data_ptr<double>(),
+                        (void*)stream);
+  else
+    splatt_hip_gram_f32(A.data_ptr<float>(), n, F, G.data_ptr<float>(),
+                        (void*)stream);
 }
 
 // key + per-other-level (idx, mat) pairs; nnz products folded by key runs
@@ -483,5 +498,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_mttkrp3", &py_gpu_mttkrp3, "3-mode CSF MTTKRP HIP kernels");
   m.def("gpu_mttkrp_flat", &py_gpu_mttkrp_flat,
         "flat expanded-CSF MTTKRP HIP kernel (3..5 modes)");
+  m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
   m.def("hip_arch", []() { return splatt_hip_kernels_arch(); });
 }

@@ -22,6 +22,7 @@ import torch
 from splatt_amd._ext import native
 from splatt_amd.csf import CsfSet, csf_alloc
 from splatt_amd.mttkrp import mttkrp
+from splatt_amd.ops.dense import gram
 from splatt_amd.sptensor import SpTensor
 
 
@@ -74,7 +75,7 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
 
     factors = [seeded_init(dims[m], rank, m, opts.seed, dtype=dtype).to(dev)
                for m in range(nm)]
-    grams = [f.T @ f for f in factors]
+    grams = [gram(f) for f in factors]
     norm_x = float(cs.csfs[0].vals.double().square().sum())
     lam = torch.ones(rank, dtype=dtype, device=dev)
     buf = torch.empty(max(dims), rank, dtype=dtype, device=dev)
@@ -100,7 +101,7 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             A = mb @ torch.cholesky_inverse(L)
             lam = _normalize(A, it)
             factors[m] = A
-            grams[m] = A.T @ A
+            grams[m] = gram(A)
 
         # fit from last mode's pre-solve MTTKRP output (reference trick):
         # <X,K> = sum_f lam_f * sum_i buf[i,f] * A_last[i,f]

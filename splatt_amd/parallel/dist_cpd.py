@@ -32,6 +32,7 @@ import torch.distributed as dist
 from splatt_amd.cpd import CpdOptions, Kruskal, seeded_init, _normalize
 from splatt_amd.csf import CsfSet, build_csf, order_modes
 from splatt_amd.mttkrp import mttkrp
+from splatt_amd.ops.dense import gram
 from splatt_amd.sptensor import SpTensor
 
 
@@ -128,7 +129,7 @@ def dist_cpd_init(shard_cs: CsfSet, part_mode: int, row0: int,
 
     grams = []
     for m in range(nm):
-        g = factors[m].T @ factors[m]
+        g = gram(factors[m])
         if m == part_mode:
             _all_reduce(g)
         grams.append(g)
@@ -185,7 +186,7 @@ def dist_cpd_step(st: DistCpdState, it: int) -> float:
             lam = _normalize(A, it)
         st.lam = lam
         st.factors[m] = A
-        g = A.T @ A
+        g = gram(A)
         if m == q:
             _all_reduce(g)
         st.grams[m] = g

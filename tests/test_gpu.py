@@ -106,3 +106,12 @@ def test_gpu_atomic_heavy_small_dims():
         out = sp.mttkrp(cs, mats_g, mode)
         ref = sp.mttkrp_stream(t, mats_c, mode)
         assert (out.cpu() - ref).abs().max().item() < 1e-7
+
+
+@pytest.mark.parametrize("rank", [8, 16, 32, 64])
+def test_gpu_gram_matches_blas(rank):
+    from splatt_amd.ops.dense import gram
+    A = torch.rand(29818, rank, dtype=torch.float64).cuda()
+    G = gram(A)
+    ref = (A.T @ A)
+    assert (G - ref).abs().max().item() < 1e-8
