@@ -41,14 +41,6 @@ __device__ __forceinline__ T ldnt(const T * p) {
   return __builtin_nontemporal_load(p);
 }
 
-// The dispatcher places block b on XCD b%8; remap so each XCD runs a
-// CONTIGUOUS block range -> its private L2 sees a narrow window of the
-// sorted factor rows instead of the whole matrix (bijective for any n).
-__device__ __forceinline__ uint32_t xcd_contig_block(uint32_t b, uint32_t n) {
-  const uint32_t xcd = b % 8u, q = n / 8u, r = n % 8u;
-  return (xcd < r ? xcd * (q + 1u) : r * (q + 1u) + (xcd - r) * q) + b / 8u;
-}
-
 // ---------------------------------------------------------- spec kernels
 // F lanes per column group, R = 64/F groups each walking a contiguous
 // sub-span; U-deep unroll for memory-level parallelism.
@@ -65,8 +57,8 @@ mttkrp_flat_kern(const int32_t * __restrict__ key,
                  V * __restrict__ out) {
   constexpr int R = WAVE / F;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t wid = (int64_t)xcd_contig_block(blockIdx.x, gridDim.x)
-                      * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + (threadIdx.x / WAVE);
   const int c = lane % F;
   const int g = lane / F;
   const int64_t w0 = wid * span;
@@ -148,8 +140,8 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
   constexpr int R = WAVE / F;
   constexpr int GB = (F >= 8) ? 8 : F;   // gather sub-batch
   const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t wid = (int64_t)xcd_contig_block(blockIdx.x, gridDim.x)
-                      * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+  const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + (threadIdx.x / WAVE);
   const int c = lane % F;
   const int g = lane / F;
   const int gbase = g * F;
