@@ -1,0 +1,82 @@
+"""MTTKRP algorithm benchmark harness.
+
+Capability parity: reference src/bench.c + cmds/cmd_bench.c — `splatt
+bench -a {splatt,csf,giga,ttbox,coord}` times each algorithm per mode per
+iteration with optional cross-validation against the gold result. Our
+algorithm set: 'stream' (COO oracle), 'csf' (hierarchical fiber-walk HIP
+kernels / C++ CPU walker), 'flat' (staged expanded-CSF HIP kernel — the
+production path).
+"""
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from splatt_amd.cpd import seeded_init
+from splatt_amd.csf import csf_alloc
+from splatt_amd.mttkrp import mttkrp, mttkrp_stream
+from splatt_amd.sptensor import SpTensor
+
+ALGS = ("flat", "csf", "stream")
+
+
+def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
+                 niters: int = 3, device: str = "cpu",
+                 policy: str = "all", validate: bool = False,
+                 seed: int = 123) -> Dict[str, dict]:
+    """Time every algorithm for every mode; returns per-alg results with
+    seconds per mode and effective GFLOP/s (3*nnz*rank flops/MTTKRP)."""
+    algs = list(algs or ALGS)
+    dev = torch.device(device)
+    mats = [seeded_init(d, rank, m, seed).to(dev)
+            for m, d in enumerate(t.dims)]
+    td = t.to(dev)
+    cs = csf_alloc(td, policy)
+    gold = None
+    if validate:
+        gold = [mttkrp_stream(t, [m.cpu() for m in mats], mode)
+                for mode in range(t.nmodes)]
+
+    def sync():
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+
+    results: Dict[str, dict] = {}
+    flops = 3.0 * t.nnz * rank
+    for alg in algs:
+        per_mode = []
+        ok = True
+        for mode in range(t.nmodes):
+            def run():
+                if alg == "stream":
+                    return mttkrp_stream(t, [m.cpu() for m in mats], mode)
+                return mttkrp(cs, mats, mode, alg=alg)
+            out = run()  # warmup + result for validation
+            if gold is not None:
+                err = (out.cpu().double() - gold[mode].double()).abs().max()
+                ok &= bool(err < 1e-6)
+            sync()
+            tic = time.perf_counter()
+            for _ in range(niters):
+                out = run()
+            sync()
+            per_mode.append((time.perf_counter() - tic) / niters)
+        results[alg] = {
+            "seconds_per_mode": per_mode,
+            "gflops_per_mode": [flops / s / 1e9 for s in per_mode],
+            "validated": ok if gold is not None else None,
+        }
+    return results
+
+
+def format_bench(results: Dict[str, dict]) -> str:
+    lines = []
+    for alg, r in results.items():
+        secs = " ".join(f"{s * 1e3:9.3f}ms" for s in r["seconds_per_mode"])
+        gfs = " ".join(f"{g:8.1f}" for g in r["gflops_per_mode"])
+        v = "" if r["validated"] is None else \
+            ("  [validated]" if r["validated"] else "  [MISMATCH]")
+        lines.append(f"  {alg:8s} per-mode: {secs}   GFLOP/s: {gfs}{v}")
+    return "\n".join(lines)

@@ -1,0 +1,112 @@
+"""CLI + auxiliary subsystems: reorder, graph/hypergraph, tiling, timers,
+bench harness (reference tests: reorder_test.c, graph.c, tile_*_test.c)."""
+import torch
+
+import splatt_amd as sp
+from splatt_amd import reorder as ro
+from splatt_amd.cli import main as cli_main
+from splatt_amd.graph import (graph_mpartite, graph_write, hgraph_nnz,
+                              hgraph_uncut, hgraph_write, part_read)
+from splatt_amd.tile import densetile, next_tileid_in_layer, tile_coords, tile_id
+
+
+def dense_of(t):
+    d = torch.zeros(*t.dims, dtype=torch.float64)
+    d.index_put_(tuple(t.inds), t.vals.double(), accumulate=True)
+    return d
+
+
+def test_perm_roundtrip(small3):
+    perm = ro.perm_rand(small3.dims, seed=3)
+    assert perm.is_valid()
+    t2 = ro.perm_apply(small3, perm)
+    # permuted dense tensor == dense tensor indexed by perms
+    a = dense_of(small3)
+    b = dense_of(t2)
+    assert torch.equal(a[perm.perms[0]][:, perm.perms[1]][:, :, perm.perms[2]], b)
+
+
+def test_perm_bfs_valid(small3):
+    perm = ro.perm_bfs(small3)
+    assert perm.is_valid()
+
+
+def test_perm_hgraph_valid(small3):
+    part = torch.randint(0, 4, (small3.nnz,))
+    perm = ro.perm_hgraph(small3, part)
+    assert perm.is_valid()
+
+
+def test_graphs(small3, tmp_path):
+    g = graph_mpartite(small3)
+    assert g.nvtxs == sum(small3.dims)
+    assert int(g.adj_ptr[-1]) == g.adj_ind.numel()
+    # symmetric: total weight even
+    assert int(g.adj_wgt.sum()) % 2 == 0
+    graph_write(g, tmp_path / "g.graph")
+    hg = hgraph_nnz(small3)
+    assert hg.nvtxs == small3.nnz
+    assert hg.eind.numel() == small3.nnz * small3.nmodes
+    hgraph_write(hg, tmp_path / "h.hgraph")
+    part = torch.zeros(small3.nnz, dtype=torch.int64)
+    assert hgraph_uncut(hg, part).numel() == hg.nhedges
+
+
+def test_densetile(small3):
+    tiled, tiling = densetile(small3, [4, 4, 4])
+    assert tiled.nnz == small3.nnz
+    assert int(tiling.tile_ptr[-1]) == small3.nnz
+    assert abs(float(tiled.vals.sum() - small3.vals.sum())) < 1e-9
+    # tile id math round-trips
+    for tid in range(64):
+        assert tile_id(tile_coords(tid, [4, 4, 4]), [4, 4, 4]) == tid
+    # layer traversal covers each layer exactly once (incl. prime dims)
+    dims = [3, 5, 2]
+    seen = set()
+    for layer in range(5):
+        tid = -1
+        while True:
+            tid = next_tileid_in_layer(tid, dims, 1, layer)
+            if tid < 0:
+                break
+            assert tid not in seen
+            seen.add(tid)
+    assert len(seen) == 3 * 5 * 2
+
+
+def test_cli_roundtrip(tmp_path, small3, capsys):
+    tns = str(tmp_path / "t.tns")
+    small3.save(tns)
+    assert cli_main(["stats", tns]) == 0
+    assert cli_main(["check", tns, "--fix", str(tmp_path / "f.tns"),
+                     "--compress"]) == 0
+    assert cli_main(["convert", tns, str(tmp_path / "t.bin"), "-t", "bin"]) == 0
+    assert cli_main(["reorder", tns, str(tmp_path / "r.tns"),
+                     "--type", "rand"]) == 0
+    t2 = sp.load(str(tmp_path / "t.bin"))
+    assert t2.nnz == small3.nnz
+    import os
+    os.chdir(tmp_path)
+    assert cli_main(["cpd", tns, "-r", "4", "-i", "3", "--device", "cpu",
+                     "--native"]) == 0
+    assert (tmp_path / "mode1.mat").exists()
+    assert (tmp_path / "lambda.mat").exists()
+
+
+def test_cli_bench(tmp_path):
+    t = sp.SpTensor.synthetic([30, 25, 40], 3000, seed=5)
+    tns = str(tmp_path / "b.tns")
+    t.save(tns)
+    assert cli_main(["bench", tns, "-r", "8", "-N", "1", "--device", "cpu",
+                     "-a", "csf,stream", "--validate"]) == 0
+
+
+def test_timers():
+    from splatt_amd.utils.timers import TimerRegistry
+    reg = TimerRegistry()
+    with reg.time("X"):
+        pass
+    with reg.time("X"):
+        pass
+    assert reg.timers["X"].count == 2
+    assert "X" in reg.report()
