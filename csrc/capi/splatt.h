@@ -1,0 +1,115 @@
+/**
+ * splatt.h — public C API of the MI355X-native sparse tensor
+ * factorization engine.
+ *
+ * Capability parity with SPLATT's public API (reference
+ * include/splatt.h + include/splatt/api_*.h): the same entry points —
+ * splatt_default_opts / splatt_free_opts, splatt_csf_load /
+ * splatt_csf_convert / splatt_free_csf, splatt_cpd_als /
+ * splatt_free_kruskal, splatt_mttkrp, splatt_version_* — with the same
+ * options-array calling convention (double[SPLATT_OPTION_NOPTIONS]) and
+ * the same Kruskal output contract. The CSF handle is opaque here (the
+ * rebuild's CSF is a flat per-level array set designed for GPU
+ * residency, not the reference's tile struct tree).
+ */
+#ifndef SPLATT_AMD_SPLATT_H
+#define SPLATT_AMD_SPLATT_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef uint64_t splatt_idx_t;
+typedef double splatt_val_t;
+
+#define SPLATT_MAX_NMODES 8
+
+typedef enum splatt_error_type {
+  SPLATT_SUCCESS = 0,
+  SPLATT_ERROR_BADINPUT = -1,
+  SPLATT_ERROR_NOMEMORY = -2,
+} splatt_error_type;
+
+typedef enum splatt_verbosity_type {
+  SPLATT_VERBOSITY_NONE = 0,
+  SPLATT_VERBOSITY_LOW = 1,
+  SPLATT_VERBOSITY_HIGH = 2,
+  SPLATT_VERBOSITY_MAX = 3,
+} splatt_verbosity_type;
+
+typedef enum splatt_csf_type {
+  SPLATT_CSF_ONEMODE = 0,
+  SPLATT_CSF_TWOMODE = 1,
+  SPLATT_CSF_ALLMODE = 2,
+} splatt_csf_type;
+
+typedef enum splatt_option_type {
+  SPLATT_OPTION_TOLERANCE = 0,  /* convergence tolerance (1e-5)       */
+  SPLATT_OPTION_NITER,          /* max ALS iterations (50)            */
+  SPLATT_OPTION_VERBOSITY,      /* splatt_verbosity_type (LOW)        */
+  SPLATT_OPTION_NTHREADS,       /* OpenMP threads (omp default)       */
+  SPLATT_OPTION_RANDSEED,       /* RNG seed (fixed default)           */
+  SPLATT_OPTION_CSF_ALLOC,      /* splatt_csf_type (TWOMODE)          */
+  SPLATT_OPTION_NOPTIONS
+} splatt_option_type;
+
+/** Opaque CSF tensor handle. */
+typedef struct splatt_csf splatt_csf;
+
+/** Kruskal tensor: the CPD output (parity: reference structs.h:25-44). */
+typedef struct splatt_kruskal {
+  splatt_idx_t rank;
+  splatt_val_t * factors[SPLATT_MAX_NMODES]; /* row-major dims[m] x rank */
+  splatt_val_t * lambda;
+  splatt_idx_t nmodes;
+  splatt_idx_t dims[SPLATT_MAX_NMODES];
+  double fit;
+} splatt_kruskal;
+
+/* ------------------------------------------------------------ options */
+double * splatt_default_opts(void);
+void splatt_free_opts(double * opts);
+
+/* ---------------------------------------------------------------- csf */
+/** Load a tensor file (.tns/.coo/.bin) into CSF form. `*tensors` receives
+ * an array of 1, 2 or nmodes CSF structures per SPLATT_OPTION_CSF_ALLOC. */
+int splatt_csf_load(const char * fname, splatt_idx_t * nmodes,
+                    splatt_csf ** tensors, const double * options);
+
+/** Convert an in-memory coordinate tensor to CSF. */
+int splatt_csf_convert(splatt_idx_t nmodes, splatt_idx_t nnz,
+                       splatt_idx_t ** inds, splatt_val_t * vals,
+                       splatt_csf ** tensors, const double * options);
+
+void splatt_free_csf(splatt_csf * tensors, const double * options);
+
+/* number of modes / dims accessors for the opaque handle */
+splatt_idx_t splatt_csf_nmodes(const splatt_csf * csf);
+splatt_idx_t splatt_csf_nnz(const splatt_csf * csf);
+void splatt_csf_dims(const splatt_csf * csf, splatt_idx_t * dims);
+
+/* -------------------------------------------------------- factorization */
+int splatt_cpd_als(const splatt_csf * tensors, splatt_idx_t nfactors,
+                   const double * options, splatt_kruskal * factored);
+void splatt_free_kruskal(splatt_kruskal * factored);
+
+/* ------------------------------------------------------------- kernels */
+/** One MTTKRP: mats[m] row-major dims[m] x ncolumns; result for `mode`
+ * written into matout (parity: api_kernels.h splatt_mttkrp). */
+int splatt_mttkrp(splatt_idx_t mode, splatt_idx_t ncolumns,
+                  const splatt_csf * tensors, splatt_val_t ** matrices,
+                  splatt_val_t * matout, const double * options);
+
+/* ------------------------------------------------------------- version */
+int splatt_version_major(void);
+int splatt_version_minor(void);
+int splatt_version_subminor(void);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif

@@ -50,7 +50,28 @@ def build_hip_objects():
     return objs
 
 
+def build_native_lib():
+    """libsplatt.so (C API) + the `splatt` host CLI binary — torch-free."""
+    bindir = ROOT / "bin"
+    bindir.mkdir(exist_ok=True)
+    lib = bindir / "libsplatt.so"
+    exe = bindir / "splatt"
+    srcs = [str(p) for p in CORE_SOURCES] + [str(ROOT / "csrc/capi/capi.cpp")]
+    newest = max(Path(s).stat().st_mtime
+                 for s in srcs + [str(ROOT / "csrc/capi/splatt.h"),
+                                  str(ROOT / "csrc/capi/splatt_main.cpp")])
+    if not lib.exists() or lib.stat().st_mtime < newest:
+        cxx = ["g++", "-O3", "-std=c++17", "-fPIC", "-fopenmp", "-march=native",
+               f"-I{ROOT}/csrc"]
+        print("[g++] libsplatt.so + splatt CLI", flush=True)
+        subprocess.check_call(cxx + ["-shared", "-o", str(lib)] + srcs)
+        subprocess.check_call(
+            cxx + ["-o", str(exe), str(ROOT / "csrc/capi/splatt_main.cpp"),
+                   f"-L{bindir}", "-lsplatt", f"-Wl,-rpath,{bindir}"])
+
+
 hip_objs = build_hip_objects()
+build_native_lib()
 
 ext = CppExtension(
     name="splatt_amd._C",

@@ -1,0 +1,110 @@
+"""C API (bin/libsplatt.so) and native `splatt` CLI binary tests
+(reference tests/api_test.c: opts defaults, version; CLI behavior)."""
+import ctypes
+import os
+import subprocess
+
+import pytest
+import torch
+
+import splatt_amd as sp
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+LIB = os.path.join(ROOT, "bin", "libsplatt.so")
+EXE = os.path.join(ROOT, "bin", "splatt")
+
+pytestmark = pytest.mark.skipif(not os.path.exists(LIB),
+                                reason="native lib not built")
+
+
+@pytest.fixture(scope="module")
+def lib():
+    L = ctypes.CDLL(LIB)
+    L.splatt_default_opts.restype = ctypes.POINTER(ctypes.c_double)
+    return L
+
+
+def test_version(lib):
+    assert lib.splatt_version_major() == 0
+    assert lib.splatt_version_minor() == 1
+
+
+def test_default_opts(lib):
+    o = lib.splatt_default_opts()
+    assert o[0] == pytest.approx(1e-5)   # tolerance
+    assert o[1] == 50                    # niter
+    lib.splatt_free_opts(o)
+
+
+def test_csf_load_and_cpd(lib, tmp_path, small3):
+    tns = str(tmp_path / "t.tns").encode()
+    small3.save(tns.decode())
+    o = lib.splatt_default_opts()
+    nmodes = ctypes.c_uint64()
+    csf = ctypes.c_void_p()
+    rc = lib.splatt_csf_load(tns, ctypes.byref(nmodes), ctypes.byref(csf), o)
+    assert rc == 0
+    assert nmodes.value == 3
+    assert lib.splatt_csf_nnz(csf) == small3.nnz
+
+    class Kruskal(ctypes.Structure):
+        _fields_ = [("rank", ctypes.c_uint64),
+                    ("factors", ctypes.POINTER(ctypes.c_double) * 8),
+                    ("lambda_", ctypes.POINTER(ctypes.c_double)),
+                    ("nmodes", ctypes.c_uint64),
+                    ("dims", ctypes.c_uint64 * 8),
+                    ("fit", ctypes.c_double)]
+
+    k = Kruskal()
+    o[1] = 5  # niter
+    rc = lib.splatt_cpd_als(csf, 8, o, ctypes.byref(k))
+    assert rc == 0
+    assert k.nmodes == 3
+    assert 0.0 <= k.fit < 1.0
+    # fit must match the Python CPU driver exactly (same seed/init)
+    ref = sp.cpd_als_cpu_native(small3, 8,
+                                sp.CpdOptions(max_iters=5, tolerance=1e-5))
+    assert abs(k.fit - ref.fit) < 1e-10
+    lib.splatt_free_kruskal(ctypes.byref(k))
+    lib.splatt_free_csf(csf, o)
+    lib.splatt_free_opts(o)
+
+
+def test_capi_mttkrp(lib, tmp_path, small3):
+    tns = str(tmp_path / "t.tns").encode()
+    small3.save(tns.decode())
+    o = lib.splatt_default_opts()
+    nmodes = ctypes.c_uint64()
+    csf = ctypes.c_void_p()
+    assert lib.splatt_csf_load(tns, ctypes.byref(nmodes), ctypes.byref(csf), o) == 0
+    rank = 8
+    mats = [sp.seeded_init(d, rank, m, 1).contiguous()
+            for m, d in enumerate(small3.dims)]
+    ptrs = (ctypes.POINTER(ctypes.c_double) * 3)(
+        *[ctypes.cast(m.data_ptr(), ctypes.POINTER(ctypes.c_double))
+          for m in mats])
+    out = torch.zeros(small3.dims[1], rank, dtype=torch.float64)
+    rc = lib.splatt_mttkrp(1, rank, csf, ptrs,
+                           ctypes.cast(out.data_ptr(),
+                                       ctypes.POINTER(ctypes.c_double)), o)
+    assert rc == 0
+    ref = sp.mttkrp_stream(small3, mats, 1)
+    assert (out - ref).abs().max() < 1e-10
+    lib.splatt_free_csf(csf, o)
+    lib.splatt_free_opts(o)
+
+
+def test_cli_binary(tmp_path, small3):
+    tns = str(tmp_path / "t.tns")
+    small3.save(tns)
+    r = subprocess.run([EXE, "stats", tns], capture_output=True, text=True)
+    assert r.returncode == 0 and "NNZ" in r.stdout
+    r = subprocess.run([EXE, "cpd", tns, "-r", "4", "-i", "3", "--nowrite"],
+                       capture_output=True, text=True, cwd=tmp_path)
+    assert r.returncode == 0 and "Final fit" in r.stdout
+    r = subprocess.run([EXE, "convert", tns, str(tmp_path / "t.bin")],
+                       capture_output=True, text=True)
+    assert r.returncode == 0
+    r = subprocess.run([EXE, "stats", str(tmp_path / "t.bin")],
+                       capture_output=True, text=True)
+    assert r.returncode == 0 and "NNZ" in r.stdout
