@@ -28,8 +28,8 @@ import torch
 import torch.distributed as dist
 
 import splatt_amd as sp
-from splatt_amd.parallel.dist_cpd import (
-    build_shard_csf, dist_cpd_init, dist_cpd_step, _world)
+from splatt_amd.parallel.dist_cpd import build_shard_csf
+from splatt_amd.parallel.grid import GridDecomp, grid_cpd_init, grid_cpd_step
 
 CONFIGS = {
     # name: (dims, nnz per shard, rank, concentration)
@@ -41,13 +41,10 @@ CONFIGS = {
 }
 
 
-def synth_shard(dims, nnz, part_mode, row0, nloc, seed, dtype, device):
-    """Generate this rank's layer shard directly at full local size.
-
-    Per-GPU shard shape == the named config shape (weak scaling: the global
-    tensor is N stacked layers along part_mode)."""
-    local_dims = list(dims)
-    local_dims[part_mode] = nloc
+def synth_box_shard(dec, nnz, seed, dtype, device):
+    """Generate this rank's grid-box shard directly at local size (indices
+    already chunk-local)."""
+    local_dims = list(dec.chunkn)
     gen_dev = device if nnz > 200_000_000 else "cpu"
     g = torch.Generator(device=gen_dev).manual_seed(seed)
     cols = []
@@ -71,6 +68,9 @@ def main():
     ap.add_argument("--dtype", default="f64", choices=["f64", "f32"])
     ap.add_argument("--device", default="cuda")
     ap.add_argument("--csf", default="all", choices=["one", "two", "all"])
+    ap.add_argument("--decomp", default="coarse", choices=["coarse", "medium"],
+                    help="coarse = 1D layers on the longest mode (weak "
+                         "scaling); medium = nmodes-D grid (strong scaling)")
     args = ap.parse_args()
 
     dims, nnz_shard, rank_f, conc = CONFIGS[args.config]
@@ -82,30 +82,42 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if world > 1:
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        if args.device != "cpu":
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl" if args.device != "cpu" else "gloo")
     device = torch.device(args.device if args.device == "cpu"
                           else f"cuda:{local_rank}")
 
-    # weak scaling: global partitioned-mode dim = N x shard dim
-    part_mode = max(range(len(dims)), key=lambda m: dims[m])
-    global_dims = list(dims)
-    global_dims[part_mode] *= world
-    nloc = dims[part_mode]
-    row0 = rank * nloc
-
     t0 = time.time()
-    shard = synth_shard(global_dims, nnz_shard, part_mode, row0, nloc,
-                        seed=0xB0B0 + rank, dtype=dtype, device=device)
+    if args.decomp == "coarse":
+        # weak scaling: global partitioned-mode dim = N x shard dim; one
+        # full-size layer per rank (grid = [.., world, ..])
+        part_mode = max(range(len(dims)), key=lambda m: dims[m])
+        global_dims = list(dims)
+        global_dims[part_mode] *= world
+        grid = [1] * len(dims)
+        grid[part_mode] = world
+        nnz_local = nnz_shard
+        scaling = "weak"
+    else:
+        # medium-grained: fixed global tensor over an auto nmodes-D grid
+        global_dims = list(dims)
+        grid = None
+        nnz_local = nnz_shard // world
+        scaling = "strong"
+    dec = GridDecomp.create(global_dims, grid=grid)
+    shard = synth_box_shard(dec, nnz_local, seed=0xB0B0 + rank, dtype=dtype,
+                            device=device)
     cs = build_shard_csf(shard, global_dims, args.csf).to(device)
     del shard
     if rank == 0:
-        print(f"# setup: shard nnz={nnz_shard} csf_bytes={cs.storage_bytes()}"
-              f" build_s={time.time() - t0:.1f}", file=sys.stderr, flush=True)
+        print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} "
+              f"csf_bytes={cs.storage_bytes()} build_s={time.time() - t0:.1f}",
+              file=sys.stderr, flush=True)
 
     opts = sp.CpdOptions(max_iters=args.warmup + args.steps, tolerance=0.0,
                          seed=0x5EED)
-    st = dist_cpd_init(cs, part_mode, row0, global_dims, rank_f, opts)
+    st = grid_cpd_init(cs, dec, rank_f, opts)
 
     def barrier_sync():
         if world > 1:
@@ -115,12 +127,12 @@ def main():
 
     it = 0
     for _ in range(args.warmup):
-        dist_cpd_step(st, it)
+        grid_cpd_step(st, it)
         it += 1
     barrier_sync()
     tic = time.time()
     for _ in range(args.steps):
-        dist_cpd_step(st, it)
+        grid_cpd_step(st, it)
         it += 1
     barrier_sync()
     elapsed = time.time() - tic
@@ -132,7 +144,7 @@ def main():
         elapsed = float(e.item())
 
     nmodes = len(global_dims)
-    nnz_global = nnz_shard * world
+    nnz_global = nnz_local * world
     flops = args.steps * nmodes * 3.0 * nnz_global * rank_f
     gflops = flops / elapsed / 1e9
     if rank == 0:
@@ -145,7 +157,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1e3, 3),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": scaling,
             "vs_baseline": None,
             "dtype": args.dtype,
             "data": "synthetic",
@@ -155,7 +167,7 @@ def main():
                 "nnz": nnz_global,
                 "cp_rank": rank_f,
                 "csf": args.csf,
-                "parallelism": f"layer-partition x{world} (RCCL/xGMI)",
+                "parallelism": f"{args.decomp}-grid {dec.grid} x{world} (RCCL/xGMI)",
                 "fit": round(st.fit, 6),
             },
         }

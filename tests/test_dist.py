@@ -78,3 +78,68 @@ def test_dist_fit_matches_single_process(tmp_path):
     assert tag == "fit"
     assert niters2 == k1.niters
     assert abs(fit2 - k1.fit) < 1e-8, (fit2, k1.fit)
+
+
+def test_best_grid():
+    from splatt_amd.parallel.grid import best_grid, prime_factors
+    assert prime_factors(12) == [3, 2, 2]
+    assert best_grid([100, 10, 10], 4) == [4, 1, 1]
+    g = best_grid([40, 30, 60], 4)
+    assert sorted(g) == [1, 2, 2] and g[2] == 2
+    assert best_grid([10, 10, 10], 8) == [2, 2, 2]
+
+
+def test_grid_localize():
+    from splatt_amd.parallel.grid import GridDecomp
+    t = sp.SpTensor.synthetic(DIMS, 2000, seed=9)
+    total = 0
+    for r in range(4):
+        dec = GridDecomp.create(list(DIMS), grid=[2, 1, 2], rank=r)
+        s = dec.localize(t)
+        total += s.nnz
+        for m in range(3):
+            assert s.dims[m] == dec.chunkn[m]
+            if s.nnz:
+                assert int(s.inds[m].max()) < dec.chunkn[m]
+    assert total == t.nnz
+
+
+def _grid_worker(rank, world, file_store, result_q):
+    torch.distributed.init_process_group(
+        "gloo", init_method=f"file://{file_store}", rank=rank,
+        world_size=world)
+    try:
+        from splatt_amd.parallel.grid import GridDecomp, grid_cpd_als
+        from splatt_amd.parallel.dist_cpd import build_shard_csf
+        t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+        dec = GridDecomp.create(list(DIMS), grid=[2, 1, 2])
+        shard = dec.localize(t)
+        cs = build_shard_csf(shard, list(DIMS), "two")
+        opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+        k = grid_cpd_als(cs, dec, RANK_F, opts)
+        if rank == 0:
+            result_q.put(("fit", k.fit, k.niters))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_grid_cpd_matches_single_process(tmp_path):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+    opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+    k1 = sp.cpd_als(t, RANK_F, opts)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    store = str(tmp_path / "store_grid")
+    procs = [ctx.Process(target=_grid_worker, args=(r, 4, store, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    tag, fit4, niters4 = q.get()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    assert niters4 == k1.niters
+    assert abs(fit4 - k1.fit) < 1e-8, (fit4, k1.fit)
