@@ -34,6 +34,11 @@ class CpdOptions:
     csf_alloc: str = "two"
     nthreads: int = 0
     verbose: bool = False
+    # per-iteration factor checkpointing (the reference has none —
+    # SURVEY.md §5 flags this as a cheap rebuild improvement)
+    checkpoint_path: str = ""
+    checkpoint_every: int = 1
+    resume: bool = False
 
 
 @dataclass
@@ -85,7 +90,20 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
     niters = 0
     ones = torch.ones(rank, rank, dtype=dtype, device=dev)
 
-    for it in range(opts.max_iters):
+    it0 = 0
+    if opts.resume and opts.checkpoint_path:
+        import os
+        if os.path.exists(opts.checkpoint_path):
+            ck = torch.load(opts.checkpoint_path, map_location=dev,
+                            weights_only=True)
+            factors = [f.to(dev) for f in ck["factors"]]
+            grams = [gram(f) for f in factors]
+            lam = ck["lambda"].to(dev)
+            it0 = int(ck["iteration"]) + 1
+            fit = old_fit = float(ck["fit"])
+            trace = list(ck.get("fit_trace", []))
+
+    for it in range(it0, opts.max_iters):
         for m in range(nm):
             mb = buf[: dims[m]]
             mttkrp(cs, factors, m, out=mb, nthreads=opts.nthreads)
@@ -118,6 +136,13 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
         niters = it + 1
         if opts.verbose:
             print(f"  its = {it + 1} fit = {fit:.5f} delta = {fit - old_fit:+.4e}")
+        if opts.checkpoint_path and (it + 1) % opts.checkpoint_every == 0:
+            import os
+            tmp = opts.checkpoint_path + ".tmp"
+            torch.save({"factors": [f.cpu() for f in factors],
+                        "lambda": lam.cpu(), "iteration": it, "fit": fit,
+                        "fit_trace": trace}, tmp)
+            os.replace(tmp, opts.checkpoint_path)
         if it > 0 and abs(fit - old_fit) < opts.tolerance:
             break
         old_fit = fit

@@ -52,3 +52,16 @@ def test_seeded_init_deterministic():
     assert torch.equal(a, b)
     assert not torch.equal(a, c)
     assert float(a.min()) >= 0.0 and float(a.max()) < 1.0
+
+
+def test_checkpoint_resume(tmp_path, small3):
+    ck = str(tmp_path / "cpd.ckpt")
+    full = sp.cpd_als(small3, 8, sp.CpdOptions(max_iters=6, tolerance=0.0))
+    # run 3 iterations with checkpointing, then resume for 3 more
+    sp.cpd_als(small3, 8, sp.CpdOptions(max_iters=3, tolerance=0.0,
+                                        checkpoint_path=ck))
+    resumed = sp.cpd_als(small3, 8, sp.CpdOptions(max_iters=6, tolerance=0.0,
+                                                  checkpoint_path=ck,
+                                                  resume=True))
+    assert abs(resumed.fit - full.fit) < 1e-10
+    assert resumed.niters == 6
