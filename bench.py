@@ -110,6 +110,9 @@ def main():
                             device=device)
     cs = build_shard_csf(shard, global_dims, args.csf).to(device)
     del shard
+    if device.type == "cuda":
+        for c in cs.csfs:
+            c.freeze_flat()   # flat kernels only need expansions + vals
     if rank == 0:
         print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} "
               f"csf_bytes={cs.storage_bytes()} build_s={time.time() - t0:.1f}",

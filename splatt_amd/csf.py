@@ -98,6 +98,16 @@ class Csf:
         cache[level] = t
         return t
 
+    def freeze_flat(self) -> "Csf":
+        """Materialize every per-level expansion and drop the fptr tree —
+        the flat MTTKRP kernels only need expansions + vals. Halves the
+        device footprint of billion-nnz ALLMODE sets (fptr is int64 and
+        ~nnz-long when fibers are short)."""
+        for l in range(self.nmodes):
+            self.ancestor_expand(l)
+        self.fptr = [None] * self.nmodes
+        return self
+
     def to_dict(self) -> dict:
         return {
             "fptr": [t.cpu() if t is not None else None for t in self.fptr],

@@ -79,3 +79,13 @@ def test_ancestor_expand_recovers_sorted_coords(small3, med4):
             exp = c.ancestor_expand(lv)
             want = keys[lv][order].to(torch.int32)
             assert torch.equal(exp, want), lv
+
+
+def test_freeze_flat_keeps_expansions(small3):
+    perm = sp.order_modes(small3.dims, "smallfirst")
+    c = build_csf(small3, perm)
+    want = [c.ancestor_expand(l).clone() for l in range(3)]
+    c.freeze_flat()
+    assert all(fp is None for fp in c.fptr)
+    for l in range(3):
+        assert torch.equal(c.ancestor_expand(l), want[l])
