@@ -24,6 +24,9 @@ import os
 import sys
 import time
 
+# avoid fragmentation OOM on billion-nnz builds (must precede torch import)
+os.environ.setdefault("PYTORCH_HIP_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 import torch.distributed as dist
 
@@ -108,11 +111,9 @@ def main():
     dec = GridDecomp.create(global_dims, grid=grid)
     shard = synth_box_shard(dec, nnz_local, seed=0xB0B0 + rank, dtype=dtype,
                             device=device)
-    cs = build_shard_csf(shard, global_dims, args.csf).to(device)
+    cs = build_shard_csf(shard, global_dims, args.csf,
+                         flat_only=device.type == "cuda")
     del shard
-    if device.type == "cuda":
-        for c in cs.csfs:
-            c.freeze_flat()   # flat kernels only need expansions + vals
     if rank == 0:
         print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} "
               f"csf_bytes={cs.storage_bytes()} build_s={time.time() - t0:.1f}",

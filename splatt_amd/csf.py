@@ -124,10 +124,13 @@ def order_modes(dims: Sequence[int], policy: str, mode: int = 0) -> List[int]:
     return [int(x) for x in native().order_modes(list(dims), policy, mode)]
 
 
-def build_csf(t: SpTensor, perm: Sequence[int]) -> Csf:
-    """Build one CSF with level->mode permutation `perm`."""
+def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False) -> Csf:
+    """Build one CSF with level->mode permutation `perm`. `flat_only`
+    (device builds): skip the fptr/fids tree — the sorted columns ARE the
+    flat kernel's expansions, so billion-nnz ALLMODE sets build ~2x faster
+    at half the transient footprint."""
     if t.device.type == "cuda":
-        return _build_csf_device(t, list(perm))
+        return _build_csf_device(t, list(perm), flat_only)
     d = native().csf_build(t.inds, t.vals, list(t.dims), list(perm))
     return Csf(dims=[int(x) for x in d["dims"]],
                dim_perm=[int(x) for x in d["dim_perm"]],
@@ -136,7 +139,8 @@ def build_csf(t: SpTensor, perm: Sequence[int]) -> Csf:
                vals=d["vals"])
 
 
-def _build_csf_device(t: SpTensor, perm: List[int]) -> Csf:
+def _build_csf_device(t: SpTensor, perm: List[int],
+                      flat_only: bool = False) -> Csf:
     """All-device CSF construction with torch/rocPRIM primitives."""
     nm, nnz = t.nmodes, t.nnz
     dev = t.device
@@ -147,6 +151,14 @@ def _build_csf_device(t: SpTensor, perm: List[int]) -> Csf:
         order = order.index_select(0, torch.argsort(keys, stable=True))
     sinds = [t.inds[perm[l]].index_select(0, order) for l in range(nm)]
     svals = t.vals.index_select(0, order)
+
+    if flat_only:
+        c = Csf(dims=list(t.dims), dim_perm=list(perm),
+                fptr=[None] * nm, fids=[None] * nm, vals=svals)
+        c.fids[nm - 1] = sinds[nm - 1].to(torch.int32)
+        cache = {l: sinds[l].to(torch.int32) for l in range(nm)}
+        object.__setattr__(c, "_expand_cache", cache)
+        return c
 
     # new-node flags per level: node at level l starts where any of levels
     # 0..l changes (diff-level trick, same invariant as the C++ builder)

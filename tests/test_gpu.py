@@ -115,3 +115,17 @@ def test_gpu_gram_matches_blas(rank):
     G = gram(A)
     ref = (A.T @ A)
     assert (G - ref).abs().max().item() < 1e-8
+
+
+def test_gpu_flat_only_build_matches(t3):
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    mats_c = make_mats(t3.dims, 16)
+    mats_g = [m.cuda() for m in mats_c]
+    td = t3.to("cuda")
+    cs = build_shard_csf(td, list(t3.dims), "all", flat_only=True)
+    for c in cs.csfs:
+        assert all(fp is None for fp in c.fptr)
+    for mode in range(3):
+        out = sp.mttkrp(cs, mats_g, mode)
+        ref = sp.mttkrp_stream(t3, mats_c, mode)
+        assert (out.cpu() - ref).abs().max().item() < 1e-8
