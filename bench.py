@@ -71,6 +71,9 @@ def main():
     ap.add_argument("--dtype", default="f64", choices=["f64", "f32"])
     ap.add_argument("--device", default="cuda")
     ap.add_argument("--csf", default="all", choices=["one", "two", "all"])
+    ap.add_argument("--gather-tiles", type=int, default=-1,
+                    help="dense-tiling buckets for factor-row gathers "
+                         "(-1 auto: target <=96MB per phase, 0 off)")
     ap.add_argument("--decomp", default="coarse", choices=["coarse", "medium"],
                     help="coarse = 1D layers on the longest mode (weak "
                          "scaling); medium = nmodes-D grid (strong scaling)")
@@ -111,11 +114,17 @@ def main():
     dec = GridDecomp.create(global_dims, grid=grid)
     shard = synth_box_shard(dec, nnz_local, seed=0xB0B0 + rank, dtype=dtype,
                             device=device)
+    gt = args.gather_tiles
+    if gt < 0:  # auto: biggest factor chunk vs ~96MB of L3 per phase
+        vbytes = 8 if args.dtype == "f64" else 4
+        big = max(dec.chunkn) * rank_f * vbytes
+        gt = min(32, max(1, (big + 96_000_000 - 1) // 96_000_000))
     cs = build_shard_csf(shard, global_dims, args.csf,
-                         flat_only=device.type == "cuda")
+                         flat_only=device.type == "cuda",
+                         gather_tiles=gt if device.type == "cuda" else 0)
     del shard
     if rank == 0:
-        print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} "
+        print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} gt={gt} "
               f"csf_bytes={cs.storage_bytes()} build_s={time.time() - t0:.1f}",
               file=sys.stderr, flush=True)
 

@@ -124,13 +124,20 @@ def order_modes(dims: Sequence[int], policy: str, mode: int = 0) -> List[int]:
     return [int(x) for x in native().order_modes(list(dims), policy, mode)]
 
 
-def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False) -> Csf:
+def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
+              gather_tiles: int = 0) -> Csf:
     """Build one CSF with level->mode permutation `perm`. `flat_only`
     (device builds): skip the fptr/fids tree — the sorted columns ARE the
     flat kernel's expansions, so billion-nnz ALLMODE sets build ~2x faster
-    at half the transient footprint."""
+    at half the transient footprint. `gather_tiles` > 1 (flat_only builds):
+    the dense-tiling analog for the flat kernel — nonzeros are bucketed by
+    the row range of the LARGEST non-root mode (outermost sort key), so
+    each phase's random factor-row gathers hit a 1/T-sized working set
+    (reference tt_densetile's cache story, tile.c:262, recast for the
+    per-XCD L2/L3 hierarchy). Output-key runs stay contiguous per bucket,
+    so the kernel is unchanged (one atomic per key run per bucket)."""
     if t.device.type == "cuda":
-        return _build_csf_device(t, list(perm), flat_only)
+        return _build_csf_device(t, list(perm), flat_only, gather_tiles)
     d = native().csf_build(t.inds, t.vals, list(t.dims), list(perm))
     return Csf(dims=[int(x) for x in d["dims"]],
                dim_perm=[int(x) for x in d["dim_perm"]],
@@ -140,7 +147,8 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False) -> Csf:
 
 
 def _build_csf_device(t: SpTensor, perm: List[int],
-                      flat_only: bool = False) -> Csf:
+                      flat_only: bool = False,
+                      gather_tiles: int = 0) -> Csf:
     """All-device CSF construction with torch/rocPRIM primitives."""
     nm, nnz = t.nmodes, t.nnz
     dev = t.device
@@ -149,6 +157,13 @@ def _build_csf_device(t: SpTensor, perm: List[int],
     for level in reversed(range(nm)):
         keys = t.inds[perm[level]].index_select(0, order)
         order = order.index_select(0, torch.argsort(keys, stable=True))
+    if flat_only and gather_tiles > 1:
+        # outermost: bucket of the largest non-root mode's row index
+        big = max(range(1, nm), key=lambda l: t.dims[perm[l]])
+        chunk = (t.dims[perm[big]] + gather_tiles - 1) // gather_tiles
+        bucket = torch.div(t.inds[perm[big]].index_select(0, order), chunk,
+                           rounding_mode="floor")
+        order = order.index_select(0, torch.argsort(bucket, stable=True))
     sinds = [t.inds[perm[l]].index_select(0, order) for l in range(nm)]
     svals = t.vals.index_select(0, order)
 

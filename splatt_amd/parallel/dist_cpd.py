@@ -66,20 +66,21 @@ def localize_shard(t: SpTensor, part_mode: int, row0: int, nloc: int) -> SpTenso
 
 
 def build_shard_csf(shard: SpTensor, global_dims: List[int],
-                    policy: str = "two", flat_only: bool = False) -> CsfSet:
+                    policy: str = "two", flat_only: bool = False,
+                    gather_tiles: int = 0) -> CsfSet:
     """CSF for a shard with mode ORDER decided by the global dims, so every
     rank picks the same kernel dispatch (root/intl/leaf) per mode."""
     nm = shard.nmodes
     if policy == "one":
         perm = order_modes(global_dims, "smallfirst")
-        c = build_csf(shard, perm, flat_only)
+        c = build_csf(shard, perm, flat_only, gather_tiles)
         return CsfSet([c], [0] * nm, [c.level_of_mode(m) for m in range(nm)])
     if policy == "two":
         perm = order_modes(global_dims, "smallfirst")
         longest = perm[-1]
-        c0 = build_csf(shard, perm, flat_only)
+        c0 = build_csf(shard, perm, flat_only, gather_tiles)
         c1 = build_csf(shard, order_modes(global_dims, "root", longest),
-                       flat_only)
+                       flat_only, gather_tiles)
         mode_csf, mode_depth = [], []
         for m in range(nm):
             if m == longest:
@@ -89,7 +90,8 @@ def build_shard_csf(shard: SpTensor, global_dims: List[int],
                 mode_csf.append(0)
                 mode_depth.append(c0.level_of_mode(m))
         return CsfSet([c0, c1], mode_csf, mode_depth)
-    csfs = [build_csf(shard, order_modes(global_dims, "root", m), flat_only)
+    csfs = [build_csf(shard, order_modes(global_dims, "root", m), flat_only,
+                      gather_tiles)
             for m in range(nm)]
     return CsfSet(csfs, list(range(nm)), [0] * nm)
 

@@ -129,3 +129,15 @@ def test_gpu_flat_only_build_matches(t3):
         out = sp.mttkrp(cs, mats_g, mode)
         ref = sp.mttkrp_stream(t3, mats_c, mode)
         assert (out.cpu() - ref).abs().max().item() < 1e-8
+
+
+def test_gpu_gather_tiled_build_matches(t3):
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    mats_c = make_mats(t3.dims, 16)
+    mats_g = [m.cuda() for m in mats_c]
+    cs = build_shard_csf(t3.to("cuda"), list(t3.dims), "all",
+                         flat_only=True, gather_tiles=4)
+    for mode in range(3):
+        out = sp.mttkrp(cs, mats_g, mode)
+        ref = sp.mttkrp_stream(t3, mats_c, mode)
+        assert (out.cpu() - ref).abs().max().item() < 1e-8
