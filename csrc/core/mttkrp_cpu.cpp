@@ -1,4 +1,5 @@
 #include "mttkrp_cpu.hpp"
+#include "partition.hpp"
 #include <cstring>
 #if defined(_OPENMP)
 #include <omp.h>
@@ -124,12 +125,36 @@ void mttkrp_csf_cpu(const Csf<V> & c, V const * const * mats,
   CsfWalker<V> w{c, mats, out, rank, outdepth, leaf};
 
   if (outdepth == 0) {
-    // root output: no write conflicts across root nodes
+    // root output: no write conflicts across root nodes. Load balance via
+    // CCP over per-root-subtree nnz (reference csf_partition_1d,
+    // csf.c:854-872) instead of dynamic chunking.
+    std::vector<int64_t> weights(nroot);
+    {
+      // nnz span per root node: compose fptr chains to the leaf level
+      std::vector<int64_t> nnzstart(c.fptr[0]);
+      for (int l = 1; l < nm - 1; ++l) {
+        #pragma omp parallel for schedule(static)
+        for (int64_t k = 0; k < (int64_t)nnzstart.size(); ++k)
+          nnzstart[k] = c.fptr[l][nnzstart[k]];
+      }
+      #pragma omp parallel for schedule(static)
+      for (int64_t s = 0; s < nroot; ++s)
+        weights[s] = nnzstart[s + 1] - nnzstart[s];
+    }
+    int nt = 1;
+#if defined(_OPENMP)
+    nt = omp_get_max_threads();
+#endif
+    const auto parts = partition_weighted(weights.data(), nroot, nt);
     #pragma omp parallel
     {
+#if defined(_OPENMP)
+      const int tid = omp_get_thread_num();
+#else
+      const int tid = 0;
+#endif
       std::vector<V> bufs((size_t)rank * (nm + 1));
-      #pragma omp for schedule(dynamic, 16)
-      for (int64_t s = 0; s < nroot; ++s) {
+      for (int64_t s = parts[tid]; s < parts[tid + 1]; ++s) {
         V * acc = bufs.data();
         for (int f = 0; f < rank; ++f) acc[f] = 0;
         w.subtree_below(0, s, acc, bufs.data() + rank);

@@ -14,6 +14,7 @@
 #include "core/matrix.hpp"
 #include "core/cpd.hpp"
 #include "core/io.hpp"
+#include "core/partition.hpp"
 
 namespace sp = splatt;
 using torch::Tensor;
@@ -499,5 +500,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_mttkrp_flat", &py_gpu_mttkrp_flat,
         "flat expanded-CSF MTTKRP HIP kernel (3..5 modes)");
   m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
+  m.def("partition_weighted", [](std::vector<int64_t> w, int nparts) {
+    int64_t bn = 0;
+    auto parts = sp::partition_weighted(w.data(), (int64_t)w.size(), nparts, &bn);
+    return py::make_tuple(parts, bn);
+  }, "optimal chains-on-chains partition (boundaries, bottleneck)");
   m.def("hip_arch", []() { return splatt_hip_kernels_arch(); });
 }

@@ -75,6 +75,26 @@ class GridDecomp:
     layer_groups: Dict[int, Optional[object]] = field(default_factory=dict)
 
     @staticmethod
+    def create_fine(global_dims: List[int],
+                    nnz_part: Optional[torch.Tensor] = None) -> "FineDecomp":
+        """Fine-grained decomposition: an arbitrary partition of the
+        NONZEROS (from an external partition file, or contiguous slices);
+        every factor is replicated and every mode's partial rows are
+        world-all-reduced (reference fine-grained mode, mpi_setup.c:181-193,
+        mpi_io.c:486-499)."""
+        world = _world()
+        r = dist.get_rank() if world > 1 else 0
+        d = FineDecomp(grid=[1] * len(global_dims), rank=r,
+                       coords=[0] * len(global_dims),
+                       chunk0=[0] * len(global_dims),
+                       chunkn=list(global_dims),
+                       global_dims=list(global_dims),
+                       nnz_part=nnz_part)
+        for m in range(len(global_dims)):
+            d.layer_groups[m] = None if world == 1 else dist.group.WORLD
+        return d
+
+    @staticmethod
     def create(global_dims: List[int], grid: Optional[List[int]] = None,
                rank: Optional[int] = None) -> "GridDecomp":
         world = _world()
@@ -142,6 +162,27 @@ class GridDecomp:
         for m in range(t.nmodes):
             inds[m] -= self.chunk0[m]
         return SpTensor(inds, t.vals[mask].clone(), list(self.chunkn))
+
+
+@dataclass
+class FineDecomp(GridDecomp):
+    """grid == [1]*nm but world > 1: factors fully replicated, partial
+    MTTKRP rows reduced across the world. repl() reflects that."""
+    nnz_part: Optional[torch.Tensor] = None
+
+    def repl(self, m: int) -> int:
+        return _world()
+
+    def localize(self, t: SpTensor) -> SpTensor:
+        world = _world()
+        if self.nnz_part is not None:
+            mask = self.nnz_part == self.rank
+        else:
+            lo, n = chunk_range(t.nnz, world, self.rank)
+            mask = torch.zeros(t.nnz, dtype=torch.bool)
+            mask[lo: lo + n] = True
+        return SpTensor(t.inds[:, mask].clone(), t.vals[mask].clone(),
+                        list(t.dims))
 
 
 def _ar(t: torch.Tensor, group=None, op=None) -> None:

@@ -110,3 +110,31 @@ def test_timers():
         pass
     assert reg.timers["X"].count == 2
     assert "X" in reg.report()
+
+
+def test_ccp_partition():
+    """Optimality properties over varied weight shapes (reference
+    thread_partition_test.c: unit/random/sorted/fibonacci x many parts)."""
+    from splatt_amd._ext import native
+    import random
+    rng = random.Random(7)
+    shapes = {
+        "unit": [1] * 53,
+        "random": [rng.randint(1, 100) for _ in range(97)],
+        "sorted": sorted(rng.randint(1, 50) for _ in range(64)),
+        "fib": [1, 1, 2, 3, 5, 8, 13, 21, 34, 55, 89, 144],
+        "spike": [1] * 30 + [1000] + [1] * 30,
+    }
+    for name, w in shapes.items():
+        for parts in (1, 2, 3, 7, 13, 31):
+            bounds, bn = native().partition_weighted(w, parts)
+            assert bounds[0] == 0 and bounds[-1] == len(w)
+            assert all(a <= b for a, b in zip(bounds, bounds[1:]))
+            # bottleneck matches the heaviest part
+            loads = [sum(w[a:b]) for a, b in zip(bounds, bounds[1:])]
+            assert max(loads) == bn, (name, parts)
+            # optimal: no feasible split with a smaller bottleneck
+            assert bn >= max(w)
+            assert bn >= (sum(w) + parts - 1) // parts
+            _, bn2 = native().partition_weighted(w, parts + 1)
+            assert bn2 <= bn

@@ -143,3 +143,46 @@ def test_grid_cpd_matches_single_process(tmp_path):
         assert p.exitcode == 0
     assert niters4 == k1.niters
     assert abs(fit4 - k1.fit) < 1e-8, (fit4, k1.fit)
+
+
+def _fine_worker(rank, world, file_store, result_q):
+    torch.distributed.init_process_group(
+        "gloo", init_method=f"file://{file_store}", rank=rank,
+        world_size=world)
+    try:
+        from splatt_amd.parallel.grid import GridDecomp, grid_cpd_als
+        from splatt_amd.parallel.dist_cpd import build_shard_csf
+        t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+        # arbitrary (hash-based) nnz partition, reference fine-grained style
+        part = (t.inds.sum(0) * 2654435761 % 2) % world
+        dec = GridDecomp.create_fine(list(DIMS), nnz_part=part)
+        shard = dec.localize(t)
+        cs = build_shard_csf(shard, list(DIMS), "two")
+        opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+        k = grid_cpd_als(cs, dec, RANK_F, opts)
+        if rank == 0:
+            result_q.put(("fit", k.fit, k.niters))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_fine_cpd_matches_single_process(tmp_path):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+    opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+    k1 = sp.cpd_als(t, RANK_F, opts)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    store = str(tmp_path / "store_fine")
+    procs = [ctx.Process(target=_fine_worker, args=(r, 2, store, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    tag, fit2, niters2 = q.get()
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert niters2 == k1.niters
+    assert abs(fit2 - k1.fit) < 1e-8, (fit2, k1.fit)
