@@ -204,6 +204,104 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
   atomic_add_g(&out[(int64_t)cur * F + c], acc);
 }
 
+// ------------------------------------------- pipelined spec kernel (v3)
+// v2 with double-buffered gather batches: batch k+1's factor-row gathers
+// are ISSUED before batch k is folded, so the fold overlaps the next
+// batch's memory latency (hipcc emits counted s_waitcnt instead of a full
+// drain per batch). Costs ~2x the batch registers.
+template <typename V, int F, int NOTHER>
+__global__ void __launch_bounds__(256)
+mttkrp_flat3_kern(const int32_t * __restrict__ key,
+                  const int32_t * __restrict__ i0,
+                  const int32_t * __restrict__ i1,
+                  const int32_t * __restrict__ i2,
+                  const int32_t * __restrict__ i3,
+                  const V * __restrict__ m0, const V * __restrict__ m1,
+                  const V * __restrict__ m2, const V * __restrict__ m3,
+                  const V * __restrict__ vals, int64_t nnz, int64_t span,
+                  V * __restrict__ out) {
+  constexpr int R = WAVE / F;
+  constexpr int GB = (F >= 8) ? 8 : F;
+  constexpr int NB = (F + GB - 1) / GB;   // batches per stream window
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + (threadIdx.x / WAVE);
+  const int c = lane % F;
+  const int g = lane / F;
+  const int gbase = g * F;
+  const int64_t w0 = wid * span;
+  if (w0 >= nnz) return;
+  const int64_t w1 = min64(nnz, w0 + span);
+  const int64_t gsz = (w1 - w0 + R - 1) / R;
+  const int64_t p0 = min64(w1, w0 + g * gsz);
+  const int64_t p1 = min64(w1, p0 + gsz);
+  if (p0 >= p1) return;
+
+  int32_t cur = key[p0];
+  V acc = (V)0;
+  int32_t kk[2][GB];
+  V vv[2][GB], a0[2][GB], a1[2][GB], a2[2][GB], a3[2][GB];
+
+  for (int64_t pb = p0; pb < p1; pb += F) {
+    const int nb = (int)min64((int64_t)F, p1 - pb);
+    const int64_t ps = pb + (c < nb ? c : nb - 1);
+    const int32_t kreg = ldnt(&key[ps]);
+    const int32_t i0reg = ldnt(&i0[ps]);
+    const int32_t i1reg = ldnt(&i1[ps]);
+    const int32_t i2reg = (NOTHER > 2) ? ldnt(&i2[ps]) : 0;
+    const int32_t i3reg = (NOTHER > 3) ? ldnt(&i3[ps]) : 0;
+    const V vreg = ldnt(&vals[ps]);
+
+    auto issue = [&](int buf, int ub, int ne) {
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        const int src = gbase + (u < ne ? ub + u : ub);
+        kk[buf][u] = __shfl(kreg, src, WAVE);
+        vv[buf][u] = __shfl(vreg, src, WAVE);
+        const int32_t j0 = __shfl(i0reg, src, WAVE);
+        const int32_t j1 = __shfl(i1reg, src, WAVE);
+        a0[buf][u] = m0[(int64_t)j0 * F + c];
+        a1[buf][u] = m1[(int64_t)j1 * F + c];
+        if (NOTHER > 2) {
+          const int32_t j2 = __shfl(i2reg, src, WAVE);
+          a2[buf][u] = m2[(int64_t)j2 * F + c];
+        }
+        if (NOTHER > 3) {
+          const int32_t j3 = __shfl(i3reg, src, WAVE);
+          a3[buf][u] = m3[(int64_t)j3 * F + c];
+        }
+      }
+    };
+    auto fold = [&](int buf, int ne) {
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        if (u >= ne) break;
+        V x = vv[buf][u] * a0[buf][u] * a1[buf][u];
+        if (NOTHER > 2) x *= a2[buf][u];
+        if (NOTHER > 3) x *= a3[buf][u];
+        if (kk[buf][u] != cur) {
+          atomic_add_g(&out[(int64_t)cur * F + c], acc);
+          acc = (V)0;
+          cur = kk[buf][u];
+        }
+        acc += x;
+      }
+    };
+
+    int ne_of[NB];
+    int nbat = 0;
+    for (int ub = 0; ub < nb; ub += GB)
+      ne_of[nbat++] = nb - ub < GB ? nb - ub : GB;
+    // issue batch b+1 before folding batch b
+    issue(0, 0, ne_of[0]);
+    for (int b = 0; b < nbat; ++b) {
+      if (b + 1 < nbat) issue((b + 1) & 1, (b + 1) * GB, ne_of[b + 1]);
+      fold(b & 1, ne_of[b]);
+    }
+  }
+  atomic_add_g(&out[(int64_t)cur * F + c], acc);
+}
+
 // ------------------------------------------------------ generic-rank kernel
 // lane = column (chunked by 64), wave walks its span serially. Correctness
 // path for ranks outside the spec set.
@@ -261,7 +359,7 @@ inline int pick_unroll() {
   // A/B lever: 0 (default) = staged v2 kernel; 4/8/16 = v1 at that unroll
   const char * e = getenv("SPLATT_MTTKRP_U");
   const int u = e ? atoi(e) : 0;
-  return (u == 4 || u == 8 || u == 16) ? u : 0;
+  return (u == 3 || u == 4 || u == 8 || u == 16) ? u : 0;
 }
 
 template <typename V>
@@ -280,9 +378,12 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
   hipLaunchKernelGGL((mttkrp_flat_kern<V, F_, N_, U_>), grid, block, 0, st, ARGS)
 #define L2K(F_, N_) \
   hipLaunchKernelGGL((mttkrp_flat2_kern<V, F_, N_>), grid, block, 0, st, ARGS)
+#define L3K(F_, N_) \
+  hipLaunchKernelGGL((mttkrp_flat3_kern<V, F_, N_>), grid, block, 0, st, ARGS)
 #define LU(F_, N_) \
   switch (uu) { case 4: L1(F_, N_, 4); break; case 16: L1(F_, N_, 16); break; \
-                case 8: L1(F_, N_, 8); break; default: L2K(F_, N_); break; }
+                case 8: L1(F_, N_, 8); break; case 3: L3K(F_, N_); break; \
+                default: L2K(F_, N_); break; }
 #define LF(N_) \
   switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
                   case 16: LU(16, N_); break; case 32: LU(32, N_); break; \
