@@ -89,10 +89,10 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if world > 1:
         if args.device != "cpu":
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group("nccl" if args.device != "cpu" else "gloo")
-    device = torch.device(args.device if args.device == "cpu"
-                          else f"cuda:{local_rank}")
+    device = torch.device(args.device if args.device == "cpu" else
+                          f"cuda:{local_rank % torch.cuda.device_count()}")
 
     t0 = time.time()
     if args.decomp == "coarse":

@@ -153,6 +153,16 @@ int splatt_mttkrp(splatt_idx_t mode, splatt_idx_t ncolumns,
   }
 }
 
+splatt_mttkrp_ws * splatt_mttkrp_alloc_ws(const splatt_csf * tensors,
+                                          splatt_idx_t, const double *) {
+  auto * ws = (splatt_mttkrp_ws*)aligned_alloc64(sizeof(splatt_mttkrp_ws));
+  ws->num_csf = (splatt_idx_t)tensors->set.csfs.size();
+  for (int m = 0; m < tensors->set.csfs[0].nmodes; ++m)
+    ws->mode_csf_map[m] = (splatt_idx_t)tensors->set.mode_csf[m];
+  return ws;
+}
+void splatt_mttkrp_free_ws(splatt_mttkrp_ws * ws) { aligned_free64(ws); }
+
 int splatt_version_major(void) { return 0; }
 int splatt_version_minor(void) { return 1; }
 int splatt_version_subminor(void) { return 0; }

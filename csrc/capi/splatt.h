@@ -103,6 +103,19 @@ int splatt_mttkrp(splatt_idx_t mode, splatt_idx_t ncolumns,
                   const splatt_csf * tensors, splatt_val_t ** matrices,
                   splatt_val_t * matout, const double * options);
 
+/** MTTKRP workspace (parity: api_kernels.h splatt_mttkrp_alloc_ws). The
+ * flat-array engine needs no scratch, so this is a thin mode->CSF
+ * dispatch record kept for API compatibility. */
+typedef struct splatt_mttkrp_ws {
+  splatt_idx_t num_csf;
+  splatt_idx_t mode_csf_map[SPLATT_MAX_NMODES];
+} splatt_mttkrp_ws;
+
+splatt_mttkrp_ws * splatt_mttkrp_alloc_ws(const splatt_csf * tensors,
+                                          splatt_idx_t ncolumns,
+                                          const double * options);
+void splatt_mttkrp_free_ws(splatt_mttkrp_ws * ws);
+
 /* ------------------------------------------------------------- version */
 int splatt_version_major(void);
 int splatt_version_minor(void);

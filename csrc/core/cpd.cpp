@@ -106,6 +106,13 @@ Kruskal<V> cpd_als(const CsfSet<V> & set, int rank, const Options & opts,
     oldfit = fit;
   }
   k.fit = fit;
+  // post-process: unit 2-norm columns, scales folded into lambda
+  // (reference cpd_post_process, cpd.c:391-411)
+  for (int m = 0; m < nm; ++m) {
+    std::vector<V> norms(F);
+    mat_normalize(k.factors[m].data(), c0.dims[m], F, norms.data(), 0);
+    for (int f = 0; f < F; ++f) k.lambda[f] *= norms[f];
+  }
   return k;
 }
 

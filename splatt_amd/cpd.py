@@ -147,8 +147,19 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             break
         old_fit = fit
 
+    _post_process(factors, lam)
     return Kruskal(factors=factors, lam=lam, fit=fit, niters=niters,
                    fit_trace=trace)
+
+
+def _post_process(factors, lam) -> None:
+    """Final renormalization: unit 2-norm columns, scales folded into
+    lambda (reference cpd_post_process, cpd.c:391-411)."""
+    for A in factors:
+        norms = A.square().sum(dim=0).sqrt()
+        norms = torch.where(norms == 0, torch.ones_like(norms), norms)
+        A /= norms
+        lam *= norms
 
 
 def cpd_als_cpu_native(t: SpTensor, rank: int,

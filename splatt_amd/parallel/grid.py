@@ -312,5 +312,15 @@ def grid_cpd_als(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
         trace.append(fit)
         if it > 0 and abs(fit - st.old_fit) < opts.tolerance:
             break
+    # post-process with GLOBAL column norms (chunked factors)
+    for m, A in enumerate(st.factors):
+        s2 = A.square().sum(dim=0)
+        if dec.grid[m] > 1:
+            s2 /= dec.repl(m)
+            _ar(s2)
+        norms = s2.sqrt()
+        norms = torch.where(norms == 0, torch.ones_like(norms), norms)
+        A /= norms
+        st.lam *= norms
     return Kruskal(factors=st.factors, lam=st.lam, fit=st.fit,
                    niters=st.niters, fit_trace=trace)
