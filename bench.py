@@ -130,6 +130,11 @@ def main():
 
     opts = sp.CpdOptions(max_iters=args.warmup + args.steps, tolerance=0.0,
                          seed=0x5EED)
+    from splatt_amd.parallel.grid import comm_stats
+    cstats = comm_stats(dec, nnz_local, rank_f,
+                        8 if args.dtype == "f64" else 4)
+    if rank == 0:
+        print(f"# comm: {cstats}", file=sys.stderr, flush=True)
     st = grid_cpd_init(cs, dec, rank_f, opts)
 
     def barrier_sync():

@@ -126,7 +126,7 @@ mttkrp_flat_kern(const int32_t * __restrict__ key,
 // and redistributes them with ds_bpermute (__shfl); the factor-row gathers
 // then issue in independent batches of 8, giving ~4x the outstanding
 // gathers per wave at lower VGPR pressure.
-template <typename V, int F, int NOTHER>
+template <typename V, int F, int NOTHER, int GBP = 8>
 __global__ void __launch_bounds__(256)
 mttkrp_flat2_kern(const int32_t * __restrict__ key,
                   const int32_t * __restrict__ i0,
@@ -138,7 +138,7 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
                   const V * __restrict__ vals, int64_t nnz, int64_t span,
                   V * __restrict__ out) {
   constexpr int R = WAVE / F;
-  constexpr int GB = (F >= 8) ? 8 : F;   // gather sub-batch
+  constexpr int GB = (F >= GBP) ? GBP : F;   // gather sub-batch
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
                       + (threadIdx.x / WAVE);
@@ -359,7 +359,7 @@ inline int pick_unroll() {
   // A/B lever: 0 (default) = staged v2 kernel; 4/8/16 = v1 at that unroll
   const char * e = getenv("SPLATT_MTTKRP_U");
   const int u = e ? atoi(e) : 0;
-  return (u == 3 || u == 4 || u == 8 || u == 16) ? u : 0;
+  return (u == 2 || u == 3 || u == 4 || u == 8 || u == 16) ? u : 0;
 }
 
 template <typename V>
@@ -378,12 +378,14 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
   hipLaunchKernelGGL((mttkrp_flat_kern<V, F_, N_, U_>), grid, block, 0, st, ARGS)
 #define L2K(F_, N_) \
   hipLaunchKernelGGL((mttkrp_flat2_kern<V, F_, N_>), grid, block, 0, st, ARGS)
+#define L2K4(F_, N_) \
+  hipLaunchKernelGGL((mttkrp_flat2_kern<V, F_, N_, 4>), grid, block, 0, st, ARGS)
 #define L3K(F_, N_) \
   hipLaunchKernelGGL((mttkrp_flat3_kern<V, F_, N_>), grid, block, 0, st, ARGS)
 #define LU(F_, N_) \
   switch (uu) { case 4: L1(F_, N_, 4); break; case 16: L1(F_, N_, 16); break; \
                 case 8: L1(F_, N_, 8); break; case 3: L3K(F_, N_); break; \
-                default: L2K(F_, N_); break; }
+                case 2: L2K4(F_, N_); break; default: L2K(F_, N_); break; }
 #define LF(N_) \
   switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
                   case 16: LU(16, N_); break; case 32: LU(32, N_); break; \

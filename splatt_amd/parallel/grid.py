@@ -302,6 +302,36 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True) -> float:
     return st.fit
 
 
+def comm_stats(dec: GridDecomp, shard_nnz: int, rank_f: int,
+               val_bytes: int = 8) -> dict:
+    """Per-rank communication volume per ALS iteration + nnz balance
+    (reference mpi_rank_stats / mpi_cpd_stats, stats.c:298-465)."""
+    world = _world()
+    per_mode = []
+    total = 0
+    for m in range(len(dec.global_dims)):
+        if world > 1 and dec.repl(m) > 1:
+            # ring all-reduce over the layer: ~2x payload per member
+            payload = dec.chunkn[m] * rank_f * val_bytes
+            vol = 2 * payload * (dec.repl(m) - 1) // max(dec.repl(m), 1)
+        else:
+            vol = 0
+        per_mode.append(vol)
+        total += vol
+    nnz_t = torch.tensor([float(shard_nnz)], dtype=torch.float64)
+    if world > 1:
+        mx = nnz_t.clone()
+        dist.all_reduce(mx, op=dist.ReduceOp.MAX)
+        sm = nnz_t.clone()
+        dist.all_reduce(sm)
+        imbalance = float(mx) / (float(sm) / world) - 1.0
+    else:
+        imbalance = 0.0
+    return {"grid": dec.grid, "comm_bytes_per_iter": total,
+            "comm_bytes_per_mode": per_mode,
+            "nnz_imbalance": round(imbalance, 4)}
+
+
 def grid_cpd_als(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
                  opts: Optional[CpdOptions] = None) -> Kruskal:
     opts = opts or CpdOptions()
