@@ -302,6 +302,107 @@ mttkrp_flat3_kern(const int32_t * __restrict__ key,
   atomic_add_g(&out[(int64_t)cur * F + c], acc);
 }
 
+// ----------------------------------------- vectorized-gather kernel (v4)
+// PMC evidence (profiles/): v2 runs at TA_BUSY ~87% — the per-CU vector
+// memory ADDRESS path is the bottleneck, not bandwidth or occupancy. v4
+// halves the address count: each lane gathers 16 B (VW = 16/sizeof(V)
+// columns) per factor row, so a row of F columns needs F/VW lane
+// addresses instead of F. Lane layout: L = F/VW lanes per column group,
+// R = 64/L groups; each lane folds VW columns in registers.
+template <typename V, int F, int NOTHER>
+__global__ void __launch_bounds__(256)
+mttkrp_flat4_kern(const int32_t * __restrict__ key,
+                  const int32_t * __restrict__ i0,
+                  const int32_t * __restrict__ i1,
+                  const int32_t * __restrict__ i2,
+                  const int32_t * __restrict__ i3,
+                  const V * __restrict__ m0, const V * __restrict__ m1,
+                  const V * __restrict__ m2, const V * __restrict__ m3,
+                  const V * __restrict__ vals, int64_t nnz, int64_t span,
+                  V * __restrict__ out) {
+  constexpr int VW = 16 / sizeof(V);        // columns per lane (16B loads)
+  constexpr int L = (F + VW - 1) / VW;      // lanes per column group
+  constexpr int R = WAVE / L;               // groups per wave
+  constexpr int GB = 8;                     // gather batch (nnz)
+  using Vec = __attribute__((ext_vector_type(VW))) V;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + (threadIdx.x / WAVE);
+  const int li = lane % L;                  // lane within group
+  const int c0 = li * VW;                   // first column of this lane
+  const int g = lane / L;
+  const int gbase = g * L;
+  const int64_t w0 = wid * span;
+  if (w0 >= nnz) return;
+  const int64_t w1 = min64(nnz, w0 + span);
+  const int64_t gsz = (w1 - w0 + R - 1) / R;
+  const int64_t p0 = min64(w1, w0 + g * gsz);
+  const int64_t p1 = min64(w1, p0 + gsz);
+  if (p0 >= p1) return;
+
+  int32_t cur = key[p0];
+  V acc[VW];
+  #pragma unroll
+  for (int w = 0; w < VW; ++w) acc[w] = (V)0;
+
+  for (int64_t pb = p0; pb < p1; pb += L) {
+    const int nb = (int)min64((int64_t)L, p1 - pb);   // stream window = L
+    const int64_t ps = pb + (li < nb ? li : nb - 1);
+    const int32_t kreg = ldnt(&key[ps]);
+    const int32_t i0reg = ldnt(&i0[ps]);
+    const int32_t i1reg = ldnt(&i1[ps]);
+    const int32_t i2reg = (NOTHER > 2) ? ldnt(&i2[ps]) : 0;
+    const int32_t i3reg = (NOTHER > 3) ? ldnt(&i3[ps]) : 0;
+    const V vreg = ldnt(&vals[ps]);
+    for (int ub = 0; ub < nb; ub += GB) {
+      const int ne = nb - ub < GB ? nb - ub : GB;
+      int32_t kk[GB];
+      V vv[GB];
+      Vec a0[GB], a1[GB], a2[GB], a3[GB];
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        const int src = gbase + (u < ne ? ub + u : ub);
+        kk[u] = __shfl(kreg, src, WAVE);
+        vv[u] = __shfl(vreg, src, WAVE);
+        const int32_t j0 = __shfl(i0reg, src, WAVE);
+        const int32_t j1 = __shfl(i1reg, src, WAVE);
+        a0[u] = *reinterpret_cast<const Vec*>(&m0[(int64_t)j0 * F + c0]);
+        a1[u] = *reinterpret_cast<const Vec*>(&m1[(int64_t)j1 * F + c0]);
+        if (NOTHER > 2) {
+          const int32_t j2 = __shfl(i2reg, src, WAVE);
+          a2[u] = *reinterpret_cast<const Vec*>(&m2[(int64_t)j2 * F + c0]);
+        }
+        if (NOTHER > 3) {
+          const int32_t j3 = __shfl(i3reg, src, WAVE);
+          a3[u] = *reinterpret_cast<const Vec*>(&m3[(int64_t)j3 * F + c0]);
+        }
+      }
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        if (u >= ne) break;
+        if (kk[u] != cur) {
+          #pragma unroll
+          for (int w = 0; w < VW; ++w) {
+            atomic_add_g(&out[(int64_t)cur * F + c0 + w], acc[w]);
+            acc[w] = (V)0;
+          }
+          cur = kk[u];
+        }
+        #pragma unroll
+        for (int w = 0; w < VW; ++w) {
+          V x = vv[u] * a0[u][w] * a1[u][w];
+          if (NOTHER > 2) x *= a2[u][w];
+          if (NOTHER > 3) x *= a3[u][w];
+          acc[w] += x;
+        }
+      }
+    }
+  }
+  #pragma unroll
+  for (int w = 0; w < VW; ++w)
+    atomic_add_g(&out[(int64_t)cur * F + c0 + w], acc[w]);
+}
+
 // ------------------------------------------------------ generic-rank kernel
 // lane = column (chunked by 64), wave walks its span serially. Correctness
 // path for ranks outside the spec set.
@@ -359,7 +460,7 @@ inline int pick_unroll() {
   // A/B lever: 0 (default) = staged v2 kernel; 4/8/16 = v1 at that unroll
   const char * e = getenv("SPLATT_MTTKRP_U");
   const int u = e ? atoi(e) : 0;
-  return (u == 2 || u == 3 || u == 4 || u == 8 || u == 16) ? u : 0;
+  return (u == 1 || u == 2 || u == 3 || u == 4 || u == 8 || u == 16) ? u : 0;
 }
 
 template <typename V>
@@ -382,10 +483,13 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
   hipLaunchKernelGGL((mttkrp_flat2_kern<V, F_, N_, 4>), grid, block, 0, st, ARGS)
 #define L3K(F_, N_) \
   hipLaunchKernelGGL((mttkrp_flat3_kern<V, F_, N_>), grid, block, 0, st, ARGS)
+#define L4K(F_, N_) \
+  hipLaunchKernelGGL((mttkrp_flat4_kern<V, F_, N_>), grid, block, 0, st, ARGS)
 #define LU(F_, N_) \
   switch (uu) { case 4: L1(F_, N_, 4); break; case 16: L1(F_, N_, 16); break; \
                 case 8: L1(F_, N_, 8); break; case 3: L3K(F_, N_); break; \
-                case 2: L2K4(F_, N_); break; default: L2K(F_, N_); break; }
+                case 2: L2K4(F_, N_); break; case 1: L2K(F_, N_); break; \
+                default: L4K(F_, N_); break; }
 #define LF(N_) \
   switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
                   case 16: LU(16, N_); break; case 32: LU(32, N_); break; \
