@@ -323,7 +323,7 @@ mttkrp_flat4_kern(const int32_t * __restrict__ key,
   constexpr int VW = 16 / sizeof(V);        // columns per lane (16B loads)
   constexpr int L = (F + VW - 1) / VW;      // lanes per column group
   constexpr int R = WAVE / L;               // groups per wave
-  constexpr int GB = 8;                     // gather batch (nnz)
+  constexpr int GB = (NOTHER > 2) ? 4 : 6;  // gather batch (register budget)
   using Vec = __attribute__((ext_vector_type(VW))) V;
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
