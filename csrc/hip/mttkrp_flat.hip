@@ -460,7 +460,7 @@ inline int pick_unroll() {
   // A/B lever: 0 (default) = staged v2 kernel; 4/8/16 = v1 at that unroll
   const char * e = getenv("SPLATT_MTTKRP_U");
   const int u = e ? atoi(e) : 0;
-  return (u == 1 || u == 2 || u == 3 || u == 4 || u == 8 || u == 16) ? u : 0;
+  return (u == 2 || u == 3 || u == 4 || u == 5 || u == 8 || u == 16) ? u : 0;
 }
 
 template <typename V>
@@ -488,8 +488,8 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
 #define LU(F_, N_) \
   switch (uu) { case 4: L1(F_, N_, 4); break; case 16: L1(F_, N_, 16); break; \
                 case 8: L1(F_, N_, 8); break; case 3: L3K(F_, N_); break; \
-                case 2: L2K4(F_, N_); break; case 1: L2K(F_, N_); break; \
-                default: L4K(F_, N_); break; }
+                case 2: L2K4(F_, N_); break; case 5: L4K(F_, N_); break; \
+                default: L2K(F_, N_); break; }
 #define LF(N_) \
   switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
                   case 16: LU(16, N_); break; case 32: LU(32, N_); break; \
