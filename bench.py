@@ -114,14 +114,13 @@ def main():
     dec = GridDecomp.create(global_dims, grid=grid)
     shard = synth_box_shard(dec, nnz_local, seed=0xB0B0 + rank, dtype=dtype,
                             device=device)
-    gt = args.gather_tiles
-    if gt < 0:  # auto: biggest factor chunk vs ~96MB of L3 per phase
-        vbytes = 8 if args.dtype == "f64" else 4
-        big = max(dec.chunkn) * rank_f * vbytes
-        gt = min(32, max(1, (big + 96_000_000 - 1) // 96_000_000))
+    gt = max(args.gather_tiles, 0)
+    stage_rank = rank_f if (args.gather_tiles < 0
+                            and device.type == "cuda") else 0
     cs = build_shard_csf(shard, global_dims, args.csf,
                          flat_only=device.type == "cuda",
-                         gather_tiles=gt if device.type == "cuda" else 0)
+                         gather_tiles=gt if device.type == "cuda" else 0,
+                         stage_rank=stage_rank)
     del shard
     if rank == 0:
         print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} gt={gt} "

@@ -396,6 +396,57 @@ void splatt_hip_mttkrp_flat_f32(
 // dense kernels, csrc/hip/dense_kernels.hip
 void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
 void splatt_hip_gram_f32(const float*, int64_t, int, float*, void*);
+// LDS-staged flat kernel, csrc/hip/mttkrp_lds.hip
+void splatt_hip_mttkrp_flat5_f64(
+    const int32_t*, const int32_t*, const int32_t*, const int32_t*,
+    const int32_t*, const double*, const double*, const double*,
+    const double*, const double*, const int64_t*, const int64_t*,
+    const int32_t*, int64_t, int32_t, int32_t, double*, int, int, void*);
+void splatt_hip_mttkrp_flat5_f32(
+    const int32_t*, const int32_t*, const int32_t*, const int32_t*,
+    const int32_t*, const float*, const float*, const float*,
+    const float*, const float*, const int64_t*, const int64_t*,
+    const int32_t*, int64_t, int32_t, int32_t, float*, int, int, void*);
+}
+
+// LDS-staged variant: idx[0]/mats[0] is the bucketed level; block
+// descriptors carry (nnz range, bucket row0)
+static void py_gpu_mttkrp_flat5(Tensor key, std::vector<Tensor> idx,
+                                std::vector<Tensor> mats, Tensor vals,
+                                Tensor blk_start, Tensor blk_end,
+                                Tensor blk_row0, int64_t chunk, int64_t dim0,
+                                Tensor out, int64_t stream) {
+  const int nother = (int)idx.size();
+  TORCH_CHECK(nother >= 2 && nother <= 4);
+  const int rank = (int)mats[0].size(1);
+  const int32_t * ip[4] = {nullptr, nullptr, nullptr, nullptr};
+  for (int t = 0; t < nother; ++t) ip[t] = idx[t].data_ptr<int32_t>();
+  const int64_t nblocks = blk_start.numel();
+  if (vals.scalar_type() == torch::kFloat64) {
+    const double * mp[4] = {nullptr, nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
+    splatt_hip_mttkrp_flat5_f64(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
+                                ip[3], mp[0], mp[1], mp[2], mp[3],
+                                vals.data_ptr<double>(),
+                                blk_start.data_ptr<int64_t>(),
+                                blk_end.data_ptr<int64_t>(),
+                                blk_row0.data_ptr<int32_t>(), nblocks,
+                                (int32_t)chunk, (int32_t)dim0,
+                                out.data_ptr<double>(), rank, nother,
+                                (void*)stream);
+  } else {
+    const float * mp[4] = {nullptr, nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+    splatt_hip_mttkrp_flat5_f32(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
+                                ip[3], mp[0], mp[1], mp[2], mp[3],
+                                vals.data_ptr<float>(),
+                                blk_start.data_ptr<int64_t>(),
+                                blk_end.data_ptr<int64_t>(),
+                                blk_row0.data_ptr<int32_t>(), nblocks,
+                                (int32_t)chunk, (int32_t)dim0,
+                                out.data_ptr<float>(), rank, nother,
+                                (void*)stream);
+  }
 }
 
 // G (FxF, pre-zeroed) += A^T A for tall-skinny device A
@@ -500,6 +551,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_mttkrp_flat", &py_gpu_mttkrp_flat,
         "flat expanded-CSF MTTKRP HIP kernel (3..5 modes)");
   m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
+  m.def("gpu_mttkrp_flat5", &py_gpu_mttkrp_flat5,
+        "LDS-staged flat MTTKRP (bucketed builds, root output)");
   m.def("partition_weighted", [](std::vector<int64_t> w, int nparts) {
     int64_t bn = 0;
     auto parts = sp::partition_weighted(w.data(), (int64_t)w.size(), nparts, &bn);

@@ -125,7 +125,8 @@ def order_modes(dims: Sequence[int], policy: str, mode: int = 0) -> List[int]:
 
 
 def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
-              gather_tiles: int = 0) -> Csf:
+              gather_tiles: int = 0, stage_rank: int = 0,
+              lds_kb: int = 48) -> Csf:
     """Build one CSF with level->mode permutation `perm`. `flat_only`
     (device builds): skip the fptr/fids tree — the sorted columns ARE the
     flat kernel's expansions, so billion-nnz ALLMODE sets build ~2x faster
@@ -137,7 +138,8 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
     per-XCD L2/L3 hierarchy). Output-key runs stay contiguous per bucket,
     so the kernel is unchanged (one atomic per key run per bucket)."""
     if t.device.type == "cuda":
-        return _build_csf_device(t, list(perm), flat_only, gather_tiles)
+        return _build_csf_device(t, list(perm), flat_only, gather_tiles,
+                                 stage_rank, lds_kb)
     d = native().csf_build(t.inds, t.vals, list(t.dims), list(perm))
     return Csf(dims=[int(x) for x in d["dims"]],
                dim_perm=[int(x) for x in d["dim_perm"]],
@@ -148,7 +150,8 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
 
 def _build_csf_device(t: SpTensor, perm: List[int],
                       flat_only: bool = False,
-                      gather_tiles: int = 0) -> Csf:
+                      gather_tiles: int = 0, stage_rank: int = 0,
+                      lds_kb: int = 48) -> Csf:
     """All-device CSF construction with torch/rocPRIM primitives."""
     nm, nnz = t.nmodes, t.nnz
     dev = t.device
@@ -157,13 +160,27 @@ def _build_csf_device(t: SpTensor, perm: List[int],
     for level in reversed(range(nm)):
         keys = t.inds[perm[level]].index_select(0, order)
         order = order.index_select(0, torch.argsort(keys, stable=True))
-    if flat_only and gather_tiles > 1:
+    stage_meta = None
+    if flat_only and (gather_tiles > 1 or stage_rank > 0):
         # outermost: bucket of the largest non-root mode's row index
         big = max(range(1, nm), key=lambda l: t.dims[perm[l]])
-        chunk = (t.dims[perm[big]] + gather_tiles - 1) // gather_tiles
-        bucket = torch.div(t.inds[perm[big]].index_select(0, order), chunk,
-                           rounding_mode="floor")
-        order = order.index_select(0, torch.argsort(bucket, stable=True))
+        dim_big = t.dims[perm[big]]
+        if stage_rank > 0:
+            # LDS-staging buckets: rows-per-bucket sized to the LDS budget
+            vbytes = t.vals.element_size()
+            chunk = max(64, (lds_kb * 1024) // (stage_rank * vbytes))
+            chunk = min(chunk, dim_big)
+            tiles = (dim_big + chunk - 1) // chunk
+        else:
+            tiles = gather_tiles
+            chunk = (dim_big + tiles - 1) // tiles
+        if tiles > 1:
+            bucket = torch.div(t.inds[perm[big]].index_select(0, order),
+                               chunk, rounding_mode="floor")
+            order = order.index_select(0, torch.argsort(bucket, stable=True))
+        if stage_rank > 0:
+            stage_meta = {"level": big, "chunk": int(chunk),
+                          "nbuckets": int(tiles)}
     sinds = [t.inds[perm[l]].index_select(0, order) for l in range(nm)]
     svals = t.vals.index_select(0, order)
 
@@ -173,6 +190,8 @@ def _build_csf_device(t: SpTensor, perm: List[int],
         c.fids[nm - 1] = sinds[nm - 1].to(torch.int32)
         cache = {l: sinds[l].to(torch.int32) for l in range(nm)}
         object.__setattr__(c, "_expand_cache", cache)
+        if stage_meta is not None:
+            object.__setattr__(c, "_stage", stage_meta)
         return c
 
     # new-node flags per level: node at level l starts where any of levels

@@ -7,6 +7,7 @@ extension is REQUIRED on CUDA tensors (no eager fallback).
 """
 from __future__ import annotations
 
+import os
 from typing import List, Optional
 
 import torch
@@ -42,19 +43,80 @@ def _gpu_mttkrp_csf(c: Csf, depth: int, mats: List[torch.Tensor],
         c.vals, ma.contiguous(), mb.contiguous(), out, stream)
 
 
+def _stage_blocks(c: Csf) -> dict:
+    """Per-workgroup (nnz range, bucket row0) descriptors for the
+    LDS-staged kernel; cached on the Csf."""
+    cached = getattr(c, "_stage_blocks", None)
+    if cached is not None:
+        return cached
+    st = c._stage  # type: ignore[attr-defined]
+    lvl, chunk, tiles = st["level"], st["chunk"], st["nbuckets"]
+    dev = c.device
+    nnz = c.nnz
+    if tiles > 1:
+        bucket = torch.div(c.ancestor_expand(lvl), chunk,
+                           rounding_mode="floor").contiguous()
+        bnd = torch.searchsorted(
+            bucket, torch.arange(tiles + 1, dtype=torch.int32, device=dev),
+            right=False).cpu()
+    else:
+        bnd = torch.tensor([0, nnz])
+    tgt = max(4096, nnz // 16384)
+    starts, ends, row0s = [], [], []
+    for b in range(len(bnd) - 1):
+        s, e = int(bnd[b]), int(bnd[b + 1])
+        p = s
+        while p < e:
+            q = min(e, p + tgt)
+            starts.append(p)
+            ends.append(q)
+            row0s.append(b * chunk)
+            p = q
+    blocks = {
+        "start": torch.tensor(starts, dtype=torch.int64, device=dev),
+        "end": torch.tensor(ends, dtype=torch.int64, device=dev),
+        "row0": torch.tensor(row0s, dtype=torch.int32, device=dev),
+        "chunk": chunk,
+        "level": lvl,
+    }
+    object.__setattr__(c, "_stage_blocks", blocks)
+    return blocks
+
+
 def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
                      out: torch.Tensor) -> None:
     """Flat expanded-CSF kernel (see csrc/hip/mttkrp_flat.hip): per-nnz
-    product of the non-output modes' rows folded by runs of the output key."""
+    product of the non-output modes' rows folded by runs of the output
+    key. Root-output dispatch on bucketed builds uses the LDS-staged
+    kernel (csrc/hip/mttkrp_lds.hip) when the rank is in the spec set."""
     nm = c.nmodes
     key = c.ancestor_expand(depth)
+    rank = int(mats[0].shape[1])
+    stream = torch.cuda.current_stream().cuda_stream
+    use_lds = (depth == 0 and getattr(c, "_stage", None) is not None
+               and rank in (4, 8, 16, 32, 64)
+               and os.environ.get("SPLATT_NO_LDS") != "1")
+    if use_lds:
+        blocks = _stage_blocks(c)
+        lvl = blocks["level"]
+        idx = [c.ancestor_expand(lvl)]
+        ms = [mats[c.dim_perm[lvl]].contiguous()]
+        for l in range(nm):
+            if l in (depth, lvl):
+                continue
+            idx.append(c.ancestor_expand(l))
+            ms.append(mats[c.dim_perm[l]].contiguous())
+        native().gpu_mttkrp_flat5(
+            key, idx, ms, c.vals, blocks["start"], blocks["end"],
+            blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[lvl]],
+            out, stream)
+        return
     idx, ms = [], []
     for l in range(nm):
         if l == depth:
             continue
         idx.append(c.ancestor_expand(l))
         ms.append(mats[c.dim_perm[l]].contiguous())
-    stream = torch.cuda.current_stream().cuda_stream
     native().gpu_mttkrp_flat(key, idx, ms, c.vals, out, stream)
 
 

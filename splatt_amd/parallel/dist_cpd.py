@@ -67,20 +67,20 @@ def localize_shard(t: SpTensor, part_mode: int, row0: int, nloc: int) -> SpTenso
 
 def build_shard_csf(shard: SpTensor, global_dims: List[int],
                     policy: str = "two", flat_only: bool = False,
-                    gather_tiles: int = 0) -> CsfSet:
+                    gather_tiles: int = 0, stage_rank: int = 0) -> CsfSet:
     """CSF for a shard with mode ORDER decided by the global dims, so every
     rank picks the same kernel dispatch (root/intl/leaf) per mode."""
     nm = shard.nmodes
     if policy == "one":
         perm = order_modes(global_dims, "smallfirst")
-        c = build_csf(shard, perm, flat_only, gather_tiles)
+        c = build_csf(shard, perm, flat_only, gather_tiles, stage_rank)
         return CsfSet([c], [0] * nm, [c.level_of_mode(m) for m in range(nm)])
     if policy == "two":
         perm = order_modes(global_dims, "smallfirst")
         longest = perm[-1]
-        c0 = build_csf(shard, perm, flat_only, gather_tiles)
+        c0 = build_csf(shard, perm, flat_only, gather_tiles, stage_rank)
         c1 = build_csf(shard, order_modes(global_dims, "root", longest),
-                       flat_only, gather_tiles)
+                       flat_only, gather_tiles, stage_rank)
         mode_csf, mode_depth = [], []
         for m in range(nm):
             if m == longest:
@@ -91,7 +91,7 @@ def build_shard_csf(shard: SpTensor, global_dims: List[int],
                 mode_depth.append(c0.level_of_mode(m))
         return CsfSet([c0, c1], mode_csf, mode_depth)
     csfs = [build_csf(shard, order_modes(global_dims, "root", m), flat_only,
-                      gather_tiles)
+                      gather_tiles, stage_rank)
             for m in range(nm)]
     return CsfSet(csfs, list(range(nm)), [0] * nm)
 
