@@ -12,7 +12,7 @@ import sys
 import torch
 
 import splatt_amd as sp
-from splatt_amd.stats import bytes_str, cpd_stats, stats_csf, stats_tt
+from splatt_amd.stats import cpd_stats, stats_csf, stats_tt
 from splatt_amd.utils.timers import TIMERS
 
 
@@ -47,9 +47,7 @@ def cmd_cpd(args) -> int:
             k = sp.cpd_als(cs, args.rank, opts)
     print(f"Final fit: {k.fit:.5f}  (iterations: {k.niters})")
     if not args.nowrite:
-        from splatt_amd._ext import native
         for m, f in enumerate(k.factors):
-            native().tns_write  # noqa: B018 (ensure ext loaded)
             with open(f"mode{m + 1}.mat", "w") as fh:
                 for row in f.cpu().tolist():
                     fh.write(" ".join(f"{x:.17g}" for x in row) + "\n")

@@ -207,7 +207,6 @@ def _build_csf_device(t: SpTensor, perm: List[int],
     fids: List[Optional[torch.Tensor]] = [None] * nm
     change = torch.zeros(nnz, dtype=torch.bool, device=dev)
     change[0] = True
-    starts_prev: Optional[torch.Tensor] = None
     starts_per_level: List[torch.Tensor] = []
     for l in range(nm - 1):
         if nnz > 1:

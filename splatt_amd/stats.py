@@ -5,8 +5,6 @@ cpd_stats:226-295) and util.c bytes_str humanization.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 from splatt_amd.csf import CsfSet
 from splatt_amd.sptensor import SpTensor
 
