@@ -126,7 +126,7 @@ def order_modes(dims: Sequence[int], policy: str, mode: int = 0) -> List[int]:
 
 def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
               gather_tiles: int = 0, stage_rank: int = 0,
-              lds_kb: int = 48) -> Csf:
+              lds_kb: int = 0) -> Csf:
     """Build one CSF with level->mode permutation `perm`. `flat_only`
     (device builds): skip the fptr/fids tree — the sorted columns ARE the
     flat kernel's expansions, so billion-nnz ALLMODE sets build ~2x faster
@@ -138,6 +138,9 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
     per-XCD L2/L3 hierarchy). Output-key runs stay contiguous per bucket,
     so the kernel is unchanged (one atomic per key run per bucket)."""
     if t.device.type == "cuda":
+        if lds_kb <= 0:
+            import os
+            lds_kb = int(os.environ.get("SPLATT_LDS_KB", "32"))
         return _build_csf_device(t, list(perm), flat_only, gather_tiles,
                                  stage_rank, lds_kb)
     d = native().csf_build(t.inds, t.vals, list(t.dims), list(perm))
