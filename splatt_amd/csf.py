@@ -140,7 +140,7 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
     if t.device.type == "cuda":
         if lds_kb <= 0:
             import os
-            lds_kb = int(os.environ.get("SPLATT_LDS_KB", "32"))
+            lds_kb = int(os.environ.get("SPLATT_LDS_KB", "24"))
         return _build_csf_device(t, list(perm), flat_only, gather_tiles,
                                  stage_rank, lds_kb)
     d = native().csf_build(t.inds, t.vals, list(t.dims), list(perm))
