@@ -20,6 +20,39 @@ from splatt_amd.mttkrp import mttkrp, mttkrp_stream
 from splatt_amd.sptensor import SpTensor
 
 ALGS = ("flat", "csf", "stream")
+LEGACY_ALGS = ("giga", "ttbox")  # reference benchmark baselines
+
+
+def _mttkrp_giga(t, mats, mode):
+    """GigaTensor-style: sparse unfolding X_(m) times the materialized
+    Khatri-Rao product (reference mttkrp_giga, mttkrp.c:1604-1649)."""
+    others = [m for m in range(t.nmodes) if m != mode]
+    ncols = 1
+    for m in others:
+        ncols *= t.dims[m]
+    col = torch.zeros(t.nnz, dtype=torch.int64, device=t.device)
+    for m in others:
+        col = col * t.dims[m] + t.inds[m]
+    X = torch.sparse_coo_tensor(torch.stack([t.inds[mode], col]), t.vals,
+                                (t.dims[mode], ncols)).coalesce()
+    K = mats[others[0]]
+    for m in others[1:]:
+        K = (K.unsqueeze(1) * mats[m].unsqueeze(0)).reshape(-1, K.shape[1])
+    return torch.sparse.mm(X, K)
+
+
+def _mttkrp_ttbox(t, mats, mode):
+    """Tensor-Toolbox style: one output column at a time via elementwise
+    products + index_add (reference mttkrp_ttbox, mttkrp.c:1655-1695)."""
+    rank = mats[0].shape[1]
+    out = torch.zeros(t.dims[mode], rank, dtype=t.vals.dtype, device=t.device)
+    for f in range(rank):
+        w = t.vals.clone()
+        for m in range(t.nmodes):
+            if m != mode:
+                w *= mats[m][t.inds[m], f]
+        out[:, f].index_add_(0, t.inds[mode], w)
+    return out
 
 
 def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
@@ -52,6 +85,10 @@ def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
             def run():
                 if alg == "stream":
                     return mttkrp_stream(t, [m.cpu() for m in mats], mode)
+                if alg == "giga":
+                    return _mttkrp_giga(td, mats, mode)
+                if alg == "ttbox":
+                    return _mttkrp_ttbox(td, mats, mode)
                 return mttkrp(cs, mats, mode, alg=alg)
             out = run()  # warmup + result for validation
             if gold is not None:

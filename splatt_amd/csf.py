@@ -158,6 +158,15 @@ def _build_csf_device(t: SpTensor, perm: List[int],
     """All-device CSF construction with torch/rocPRIM primitives."""
     nm, nnz = t.nmodes, t.nnz
     dev = t.device
+    if nnz == 0:
+        c = Csf(dims=list(t.dims), dim_perm=list(perm),
+                fptr=[torch.zeros(1, dtype=torch.int64, device=dev)
+                      if l < nm - 1 else None for l in range(nm)],
+                fids=[None if l == 0 else
+                      torch.zeros(0, dtype=torch.int32, device=dev)
+                      for l in range(nm)],
+                vals=t.vals.clone())
+        return c
     # lexicographic stable sort: least-significant level first
     order = torch.arange(nnz, device=dev)
     for level in reversed(range(nm)):

@@ -354,3 +354,53 @@ def grid_cpd_als(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
         st.lam *= norms
     return Kruskal(factors=st.factors, lam=st.lam, fit=st.fit,
                    niters=st.niters, fit_trace=trace)
+
+
+def load_shard(path: str, dec: GridDecomp, dtype=None):
+    """Distributed tensor load: every rank reads the (shared-filesystem)
+    file and keeps its box (single-node analog of mpi_tt_read,
+    mpi/mpi_io.c:756; chunked root->rank streaming is unnecessary when all
+    ranks share one node's filesystem)."""
+    import torch as _torch
+    t = SpTensor.load(path, dtype or _torch.float64)
+    return dec.localize(t)
+
+
+def gather_factors(k: Kruskal, dec: GridDecomp):
+    """Assemble full (global-row) factor matrices on rank 0 from chunked
+    factors (reference mpi_write_mats root-gather, mpi/mpi_io.c:927)."""
+    world = _world()
+    if world == 1:
+        return list(k.factors)
+    rank = dist.get_rank()
+    full = []
+    for m, A in enumerate(k.factors):
+        if dec.grid[m] == 1:
+            full.append(A if rank == 0 else None)
+            continue
+        pieces = [None] * world
+        dist.all_gather_object(pieces, (dec.coords[m], dec.chunk0[m],
+                                        A.cpu()))
+        if rank == 0:
+            seen = {}
+            for coord, c0, chunk in pieces:
+                seen[c0] = chunk   # layer replicas collapse
+            rows = [seen[c0] for c0 in sorted(seen)]
+            full.append(__import__("torch").cat(rows, dim=0))
+        else:
+            full.append(None)
+    return full
+
+
+def write_factors(k: Kruskal, dec: GridDecomp, prefix: str = "") -> None:
+    """Rank 0 writes modeN.mat + lambda.mat with GLOBAL rows."""
+    full = gather_factors(k, dec)
+    if _world() > 1 and dist.get_rank() != 0:
+        return
+    for m, A in enumerate(full):
+        with open(f"{prefix}mode{m + 1}.mat", "w") as fh:
+            for row in A.cpu().tolist():
+                fh.write(" ".join(f"{x:.17g}" for x in row) + "\n")
+    with open(f"{prefix}lambda.mat", "w") as fh:
+        for x in k.lam.cpu().tolist():
+            fh.write(f"{x:.17g}\n")

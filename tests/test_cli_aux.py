@@ -138,3 +138,13 @@ def test_ccp_partition():
             assert bn >= (sum(w) + parts - 1) // parts
             _, bn2 = native().partition_weighted(w, parts + 1)
             assert bn2 <= bn
+
+
+def test_legacy_bench_algs():
+    import splatt_amd as sp
+    from splatt_amd.benchmarks import bench_mttkrp
+    t = sp.SpTensor.synthetic([25, 20, 30], 2000, seed=5)
+    res = bench_mttkrp(t, 8, ["csf", "giga", "ttbox"], 1, device="cpu",
+                       validate=True)
+    for alg in ("csf", "giga", "ttbox"):
+        assert res[alg]["validated"], alg

@@ -186,3 +186,24 @@ def test_fine_cpd_matches_single_process(tmp_path):
         assert p.exitcode == 0
     assert niters2 == k1.niters
     assert abs(fit2 - k1.fit) < 1e-8, (fit2, k1.fit)
+
+
+def test_load_shard_and_write_factors(tmp_path):
+    """Single-process path of the distributed IO helpers."""
+    from splatt_amd.parallel.grid import (GridDecomp, grid_cpd_als,
+                                          load_shard, write_factors)
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    t = sp.SpTensor.synthetic(DIMS, 1000, seed=4)
+    p = tmp_path / "t.tns"
+    t.save(p)
+    dec = GridDecomp.create(list(DIMS))
+    shard = load_shard(str(p), dec)
+    assert shard.nnz == t.nnz
+    cs = build_shard_csf(shard, list(DIMS), "two")
+    k = grid_cpd_als(cs, dec, 4, sp.CpdOptions(max_iters=2, tolerance=0.0))
+    import os
+    os.chdir(tmp_path)
+    write_factors(k, dec)
+    for m in range(3):
+        assert (tmp_path / f"mode{m + 1}.mat").exists()
+    assert (tmp_path / "lambda.mat").exists()
