@@ -152,11 +152,15 @@ def main():
     runner = None
     if world == 1 and device.type == "cuda"             and os.environ.get("SPLATT_NO_GRAPH") != "1":
         from splatt_amd.parallel.graph_exec import GraphStepRunner
-        runner = GraphStepRunner(st)
-        if not runner.capture():
-            runner = None
+        _r = GraphStepRunner(st)
+        runner = _r if _r.capture() else None
         if rank == 0:
             print(f"# exec: {'hipGraph-captured' if runner else 'eager'} step",
+                  file=sys.stderr, flush=True)
+        if runner is None and rank == 0:
+            from splatt_amd.parallel import graph_exec  # noqa
+            print(f"# graph capture error: "
+                  f"{getattr(_r, 'capture_error', None)}",
                   file=sys.stderr, flush=True)
 
     barrier_sync()

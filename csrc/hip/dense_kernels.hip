@@ -73,6 +73,92 @@ void launch_gram(const V * A, int64_t n, int F, V * G, hipStream_t st) {
     hipLaunchKernelGGL((gram_kern<V, 16>), grid, block, 0, st, A, n, F, rows_per_blk, G);
 }
 
+// spd_inverse: Ginv = G^-1 for SPD G (F x F, F <= 64) in ONE workgroup —
+// Cholesky G = L L^T, triangular inverse X = L^-1, Ginv = X^T X, all in
+// LDS. Replaces the rocSOLVER potrf/potri chain (~15 launches + a host
+// info sync that breaks hipGraph capture) for the ALS normal equations.
+// On breakdown (non-positive pivot) retries with escalating Tikhonov
+// jitter, like the host solver (csrc/core/matrix.cpp solve_normals).
+template <typename V>
+__global__ void __launch_bounds__(64)
+spd_inverse_kern(const V * __restrict__ G, V * __restrict__ Ginv, int F) {
+  // one 32KB LDS array; Ginv doubles as scratch for X = L^-1
+  __shared__ V Lm[64 * 64];
+  __shared__ V diagmax_sh;
+  __shared__ int fail_sh;
+  const int tid = threadIdx.x;
+
+  V jitter = (V)0;
+  if (tid == 0) {
+    V mx = (V)0;
+    for (int j = 0; j < F; ++j) {
+      const V d = G[j * F + j];
+      mx = mx > d ? mx : d;
+    }
+    diagmax_sh = mx;
+  }
+  __syncthreads();
+  const V diagmax = diagmax_sh;
+
+  for (int attempt = 0; attempt < 24; ++attempt) {
+    for (int e = tid; e < F * F; e += 64) Lm[e] = G[e];
+    if (tid == 0) fail_sh = 0;
+    __syncthreads();
+    if (jitter > (V)0 && tid < F) Lm[tid * F + tid] += jitter;
+    __syncthreads();
+    // Cholesky, column by column; lanes parallel over rows
+    for (int j = 0; j < F; ++j) {
+      if (tid == 0) {
+        V d = Lm[j * F + j];
+        for (int k = 0; k < j; ++k) d -= Lm[j * F + k] * Lm[j * F + k];
+        if (d <= (V)0) { fail_sh = 1; d = (V)1; }
+        Lm[j * F + j] = sqrt(d);
+      }
+      __syncthreads();
+      if (fail_sh) break;
+      const V dj = Lm[j * F + j];
+      for (int i = j + 1 + tid; i < F; i += 64) {
+        V s = Lm[i * F + j];
+        for (int k = 0; k < j; ++k) s -= Lm[i * F + k] * Lm[j * F + k];
+        Lm[i * F + j] = s / dj;
+      }
+      __syncthreads();
+    }
+    if (!fail_sh) break;
+    jitter = (jitter == (V)0) ? diagmax * (V)1e-12 : jitter * (V)100;
+    __syncthreads();
+  }
+
+  // X = L^-1 (lower): forward substitution, lanes parallel over columns.
+  // X lands in the GLOBAL output buffer (scratch), with per-lane private
+  // column traffic only — no cross-lane reads of X.
+  for (int c = tid; c < F; c += 64) {
+    V xcol[64];
+    for (int i = c; i < F; ++i) {
+      if (i == c) {
+        xcol[i] = (V)1 / Lm[i * F + i];
+      } else {
+        V s = (V)0;
+        for (int k = c; k < i; ++k) s += Lm[i * F + k] * xcol[k];
+        xcol[i] = -s / Lm[i * F + i];
+      }
+      Ginv[i * F + c] = xcol[i];
+    }
+    for (int i = 0; i < c; ++i) Ginv[i * F + c] = (V)0;
+  }
+  __syncthreads();
+  // pull X into LDS (L no longer needed), then Ginv = X^T X
+  for (int e = tid; e < F * F; e += 64) Lm[e] = Ginv[e];
+  __syncthreads();
+  for (int e = tid; e < F * F; e += 64) {
+    const int i = e / F, j = e % F;
+    const int k0 = i > j ? i : j;
+    V s = (V)0;
+    for (int k = k0; k < F; ++k) s += Lm[k * F + i] * Lm[k * F + j];
+    Ginv[e] = s;
+  }
+}
+
 }  // namespace
 
 extern "C" void splatt_hip_gram_f64(const double * A, int64_t n, int F,
@@ -82,4 +168,14 @@ extern "C" void splatt_hip_gram_f64(const double * A, int64_t n, int F,
 extern "C" void splatt_hip_gram_f32(const float * A, int64_t n, int F,
                                     float * G, void * stream) {
   launch_gram<float>(A, n, F, G, (hipStream_t)stream);
+}
+extern "C" void splatt_hip_spd_inverse_f64(const double * G, double * Ginv,
+                                           int F, void * stream) {
+  hipLaunchKernelGGL((spd_inverse_kern<double>), dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, G, Ginv, F);
+}
+extern "C" void splatt_hip_spd_inverse_f32(const float * G, float * Ginv,
+                                           int F, void * stream) {
+  hipLaunchKernelGGL((spd_inverse_kern<float>), dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, G, Ginv, F);
 }

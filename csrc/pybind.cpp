@@ -396,6 +396,8 @@ void splatt_hip_mttkrp_flat_f32(
 // dense kernels, csrc/hip/dense_kernels.hip
 void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
 void splatt_hip_gram_f32(const float*, int64_t, int, float*, void*);
+void splatt_hip_spd_inverse_f64(const double*, double*, int, void*);
+void splatt_hip_spd_inverse_f32(const float*, float*, int, void*);
 // LDS-staged flat kernel, csrc/hip/mttkrp_lds.hip
 void splatt_hip_mttkrp_flat5_f64(
     const int32_t*, const int32_t*, const int32_t*, const int32_t*,
@@ -551,6 +553,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_mttkrp_flat", &py_gpu_mttkrp_flat,
         "flat expanded-CSF MTTKRP HIP kernel (3..5 modes)");
   m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
+  m.def("gpu_spd_inverse", [](Tensor G, Tensor Ginv, int64_t stream) {
+    const int F = (int)G.size(0);
+    if (G.scalar_type() == torch::kFloat64)
+      splatt_hip_spd_inverse_f64(G.data_ptr<double>(), Ginv.data_ptr<double>(),
+                                 F, (void*)stream);
+    else
+      splatt_hip_spd_inverse_f32(G.data_ptr<float>(), Ginv.data_ptr<float>(),
+                                 F, (void*)stream);
+  }, "Ginv = G^-1 for SPD FxF (F<=64), one-workgroup Cholesky");
   m.def("gpu_mttkrp_flat5", &py_gpu_mttkrp_flat5,
         "LDS-staged flat MTTKRP (bucketed builds, root output)");
   m.def("partition_weighted", [](std::vector<int64_t> w, int nparts) {

@@ -22,7 +22,7 @@ import torch
 from splatt_amd._ext import native
 from splatt_amd.csf import CsfSet, csf_alloc
 from splatt_amd.mttkrp import mttkrp
-from splatt_amd.ops.dense import gram
+from splatt_amd.ops.dense import gram, spd_inverse
 from splatt_amd.sptensor import SpTensor
 
 
@@ -112,11 +112,9 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
                 if o != m:
                     G *= grams[o]
             # solve A * G = mttkrp  =>  G^T A^T = mttkrp^T (G symmetric)
-            L = torch.linalg.cholesky(
-                G + 1e-12 * torch.eye(rank, dtype=dtype, device=dev) * G.diagonal().abs().max())
             # F x F inverse once, then one well-shaped (n x F)(F x F) GEMM —
             # beats a trsm against an n-row RHS at these tiny F
-            A = mb @ torch.cholesky_inverse(L)
+            A = mb @ spd_inverse(G)
             lam = _normalize(A, it)
             factors[m] = A
             grams[m] = gram(A)

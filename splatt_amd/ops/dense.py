@@ -16,3 +16,22 @@ def gram(A: torch.Tensor) -> torch.Tensor:
         native().gpu_gram(A, G, torch.cuda.current_stream().cuda_stream)
         return G
     return A.T @ A
+
+
+def spd_inverse(G: torch.Tensor) -> torch.Tensor:
+    """Inverse of an SPD F x F matrix. Device path (F <= 64): the
+    one-workgroup HIP Cholesky kernel — hipGraph-capture-safe, no host
+    sync, one launch instead of rocSOLVER's potrf/potri chain. Includes
+    escalating Tikhonov jitter on breakdown (reference gelss fallback
+    analog, matrix.c:554-599)."""
+    F = G.shape[0]
+    if G.device.type == "cuda" and F <= 64 and G.dtype in (torch.float64,
+                                                           torch.float32):
+        Ginv = torch.empty_like(G)
+        native().gpu_spd_inverse(G.contiguous(), Ginv,
+                                 torch.cuda.current_stream().cuda_stream)
+        return Ginv
+    L = torch.linalg.cholesky(
+        G + 1e-12 * G.diagonal().abs().max()
+        * torch.eye(F, dtype=G.dtype, device=G.device))
+    return torch.cholesky_inverse(L)

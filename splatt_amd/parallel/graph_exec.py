@@ -68,9 +68,7 @@ class GraphStepRunner:
                     first = False
                 else:
                     self.G.mul_(self.grams[o])
-            self.G.add_(self.eye)
-            torch.linalg.cholesky(self.G, out=self.L)
-            torch.cholesky_inverse(self.L, out=self.Ginv)
+            native().gpu_spd_inverse(self.G, self.Ginv, stream)
             torch.mm(mb, self.Ginv, out=self.A[m])
             # max-norm normalize (steady state)
             torch.amax(torch.abs(self.A[m]), dim=0, out=self.lam)
@@ -96,6 +94,7 @@ class GraphStepRunner:
         self.fit_parts[1] = lamd @ self.G.double() @ lamd
 
     def capture(self) -> bool:
+        self.capture_error = None
         try:
             s = torch.cuda.Stream()
             with torch.cuda.stream(s):
@@ -107,7 +106,8 @@ class GraphStepRunner:
                 self._static_step()
             self.graph = g
             return True
-        except Exception:
+        except Exception as e:  # noqa: BLE001 - fall back to eager
+            self.capture_error = repr(e)
             self.graph = None
             return False
 

@@ -27,7 +27,7 @@ import torch.distributed as dist
 from splatt_amd.cpd import CpdOptions, Kruskal, seeded_init
 from splatt_amd.csf import CsfSet
 from splatt_amd.mttkrp import mttkrp
-from splatt_amd.ops.dense import gram
+from splatt_amd.ops.dense import gram, spd_inverse
 from splatt_amd.sptensor import SpTensor
 
 
@@ -243,7 +243,6 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True) -> float:
     dev = st.buf.device
     dtype = st.buf.dtype
     F = st.factors[0].shape[1]
-    eye = torch.eye(F, dtype=dtype, device=dev)
 
     for m in range(nm):
         mb = st.buf[: dec.chunkn[m]]
@@ -257,8 +256,7 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True) -> float:
         for o in range(nm):
             if o != m:
                 G *= st.grams[o]
-        L = torch.linalg.cholesky(G + 1e-12 * G.diagonal().abs().max() * eye)
-        Ginv = torch.cholesky_inverse(L)
+        Ginv = spd_inverse(G)
         if work is not None:
             work.wait()
         A = mb @ Ginv

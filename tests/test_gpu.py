@@ -169,3 +169,37 @@ def test_gpu_lds_staged_4mode():
         out = sp.mttkrp(cs, mats_g, mode)
         ref = sp.mttkrp_stream(t, mats_c, mode)
         assert (out.cpu() - ref).abs().max().item() < 1e-8, mode
+
+
+@pytest.mark.parametrize("rank", [8, 16, 32, 64])
+def test_gpu_spd_inverse(rank):
+    from splatt_amd.ops.dense import spd_inverse
+    A = torch.rand(500, rank, dtype=torch.float64).cuda()
+    G = A.T @ A + 0.1 * torch.eye(rank, dtype=torch.float64).cuda()
+    Ginv = spd_inverse(G)
+    err = (G @ Ginv - torch.eye(rank, dtype=torch.float64).cuda()).abs().max()
+    assert float(err) < 1e-9
+
+
+def test_gpu_graph_step_matches_eager(t3):
+    """hipGraph-captured iteration == eager iteration (same fit)."""
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    from splatt_amd.parallel.grid import GridDecomp, grid_cpd_init, grid_cpd_step
+    from splatt_amd.parallel.graph_exec import GraphStepRunner
+    opts = sp.CpdOptions(max_iters=6, tolerance=0.0)
+    dec = GridDecomp.create(list(t3.dims))
+
+    cs = build_shard_csf(t3.to("cuda"), list(t3.dims), "all", flat_only=True,
+                         stage_rank=16)
+    st = grid_cpd_init(cs, dec, 16, opts)
+    grid_cpd_step(st, 0)
+    runner = GraphStepRunner(st)
+    assert runner.capture(), runner.capture_error
+    for _ in range(3):
+        runner.replay()
+    fit_graph = runner.finalize(st.norm_x)
+
+    st2 = grid_cpd_init(cs, dec, 16, opts)
+    for it in range(6):  # 1 eager + 2 warmup-in-capture + 3 replays
+        fit_eager = grid_cpd_step(st2, it)
+    assert abs(fit_graph - fit_eager) < 1e-6, (fit_graph, fit_eager)
