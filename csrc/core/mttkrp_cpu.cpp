@@ -39,6 +39,7 @@ struct CsfWalker {
   int rank;
   int outdepth;
   int leaf;
+  bool use_atomics = true;  // false when the output is thread-private
 
   // accumulate g over children of (level, node) into acc (acc zeroed by caller)
   void subtree_below(int level, int64_t node, V * acc, V * scratch) const {
@@ -75,9 +76,13 @@ struct CsfWalker {
         for (int64_t j = start; j < end; ++j) {
           const V v = c.vals[j];
           V * orow = out + (idx_t)c.fids[leaf][j] * rank;
-          for (int f = 0; f < rank; ++f) {
-            #pragma omp atomic
-            orow[f] += v * above[f];
+          if (use_atomics) {
+            for (int f = 0; f < rank; ++f) {
+              #pragma omp atomic
+              orow[f] += v * above[f];
+            }
+          } else {
+            for (int f = 0; f < rank; ++f) orow[f] += v * above[f];
           }
         }
       } else {
@@ -87,9 +92,13 @@ struct CsfWalker {
           for (int f = 0; f < rank; ++f) below[f] = 0;
           subtree_below(child_level, n, below, bufs + rank);
           V * orow = out + (idx_t)c.fids[child_level][n] * rank;
-          for (int f = 0; f < rank; ++f) {
-            #pragma omp atomic
-            orow[f] += above[f] * below[f];
+          if (use_atomics) {
+            for (int f = 0; f < rank; ++f) {
+              #pragma omp atomic
+              orow[f] += above[f] * below[f];
+            }
+          } else {
+            for (int f = 0; f < rank; ++f) orow[f] += above[f] * below[f];
           }
         }
       }
@@ -122,7 +131,7 @@ void mttkrp_csf_cpu(const Csf<V> & c, V const * const * mats,
   if (nthreads > 0) omp_set_num_threads(nthreads);
 #endif
 
-  CsfWalker<V> w{c, mats, out, rank, outdepth, leaf};
+  CsfWalker<V> w{c, mats, out, rank, outdepth, leaf, true};
 
   if (outdepth == 0) {
     // root output: no write conflicts across root nodes. Load balance via
@@ -165,15 +174,53 @@ void mttkrp_csf_cpu(const Csf<V> & c, V const * const * mats,
     }
   } else {
     const V * Mroot = mats[c.dim_perm[0]];
-    #pragma omp parallel
-    {
-      std::vector<V> bufs((size_t)rank * (nm + 2));
-      #pragma omp for schedule(dynamic, 16)
-      for (int64_t s = 0; s < nroot; ++s) {
-        const idx_t rid = root_labeled ? (idx_t)c.fids[0][s] : (idx_t)s;
-        const V * above0 = Mroot + rid * rank;
-        if (outdepth == 0) continue;  // unreachable
-        w.walk_down(0, s, above0, bufs.data());
+    int nt = 1;
+#if defined(_OPENMP)
+    nt = omp_get_max_threads();
+#endif
+    // privatization heuristic (reference p_is_privatized, mttkrp.c:221-236):
+    // replicate the output per thread when it is small relative to nnz,
+    // then tree-reduce — avoids the per-element atomics of the scatter
+    // walkers for short modes.
+    const size_t out_elems = (size_t)c.dims[mode] * rank;
+    const bool privatize =
+        nt > 1 && (double)out_elems * nt <= 0.1 * (double)c.nnz * rank;
+    if (privatize) {
+      std::vector<V> priv((size_t)nt * out_elems, (V)0);
+      #pragma omp parallel
+      {
+#if defined(_OPENMP)
+        const int tid = omp_get_thread_num();
+#else
+        const int tid = 0;
+#endif
+        CsfWalker<V> wp{c, mats, priv.data() + (size_t)tid * out_elems,
+                        rank, outdepth, leaf, /*use_atomics=*/false};
+        std::vector<V> bufs((size_t)rank * (nm + 2));
+        #pragma omp for schedule(dynamic, 16)
+        for (int64_t s = 0; s < nroot; ++s) {
+          const idx_t rid = root_labeled ? (idx_t)c.fids[0][s] : (idx_t)s;
+          wp.walk_down(0, s, Mroot + rid * rank, bufs.data());
+        }
+        // tree-style reduction: each thread sums its block across copies
+        #pragma omp barrier
+        #pragma omp for schedule(static)
+        for (int64_t e = 0; e < (int64_t)out_elems; ++e) {
+          V acc = (V)0;
+          for (int t = 0; t < nt; ++t) acc += priv[(size_t)t * out_elems + e];
+          out[e] = acc;
+        }
+      }
+    } else {
+      #pragma omp parallel
+      {
+        std::vector<V> bufs((size_t)rank * (nm + 2));
+        #pragma omp for schedule(dynamic, 16)
+        for (int64_t s = 0; s < nroot; ++s) {
+          const idx_t rid = root_labeled ? (idx_t)c.fids[0][s] : (idx_t)s;
+          const V * above0 = Mroot + rid * rank;
+          w.walk_down(0, s, above0, bufs.data());
+        }
       }
     }
   }
