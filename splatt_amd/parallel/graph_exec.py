@@ -11,12 +11,11 @@ eager. Falls back to the eager step if capture fails.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch
 
 from splatt_amd.mttkrp import mttkrp
-from splatt_amd.ops.dense import gram as gram_op
 from splatt_amd._ext import native
 
 
@@ -37,7 +36,6 @@ class GraphStepRunner:
         for m in range(nm):
             self.A[m].copy_(st.factors[m])
         self.G = torch.empty(F, F, dtype=dtype, device=dev)
-        self.L = torch.empty(F, F, dtype=dtype, device=dev)
         self.Ginv = torch.empty(F, F, dtype=dtype, device=dev)
         self.grams = [torch.empty(F, F, dtype=dtype, device=dev)
                       for _ in range(nm)]
@@ -45,13 +43,12 @@ class GraphStepRunner:
             self.grams[m].copy_(st.grams[m])
         self.lam = torch.empty(F, dtype=dtype, device=dev)
         self.lam.copy_(st.lam)
-        self.eye = torch.eye(F, dtype=dtype, device=dev) * 1e-12
         self.fit_parts = torch.zeros(2, dtype=torch.float64, device=dev)
         self.buf = st.buf
         self.tmp = torch.empty_like(st.buf)
 
     def _static_step(self):
-        st, nm, F = self.st, self.nm, self.F
+        st, nm = self.st, self.nm
         dec = st.dec
         stream = torch.cuda.current_stream().cuda_stream
         for m in range(nm):
