@@ -14,6 +14,8 @@ from splatt_amd.sptensor import SpTensor
 from splatt_amd.csf import Csf, CsfSet, CsfAllocPolicy, build_csf, csf_alloc, order_modes
 from splatt_amd.mttkrp import mttkrp, mttkrp_stream
 from splatt_amd.cpd import CpdOptions, Kruskal, cpd_als, cpd_als_cpu_native, seeded_init
+from splatt_amd.kruskal import (kruskal_fit, kruskal_innerprod, kruskal_norm,
+                                kruskal_to_dense)
 
 load = SpTensor.load
 
@@ -21,4 +23,5 @@ __all__ = [
     "SpTensor", "Csf", "CsfSet", "CsfAllocPolicy", "build_csf", "csf_alloc",
     "order_modes", "mttkrp", "mttkrp_stream", "CpdOptions", "Kruskal",
     "cpd_als", "cpd_als_cpu_native", "seeded_init", "load",
+    "kruskal_fit", "kruskal_innerprod", "kruskal_norm", "kruskal_to_dense",
 ]

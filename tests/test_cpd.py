@@ -65,3 +65,18 @@ def test_checkpoint_resume(tmp_path, small3):
                                                   resume=True))
     assert abs(resumed.fit - full.fit) < 1e-10
     assert resumed.niters == 6
+
+
+def test_kruskal_utilities(small3):
+    from splatt_amd.kruskal import (kruskal_fit, kruskal_innerprod,
+                                    kruskal_norm, kruskal_to_dense)
+    k = sp.cpd_als(small3, 6, sp.CpdOptions(max_iters=8, tolerance=0.0))
+    # from-scratch fit must match the in-loop fit closely
+    assert abs(kruskal_fit(k, small3) - k.fit) < 1e-6
+    # dense reconstruction consistency
+    dense = kruskal_to_dense(k)
+    assert abs(float(dense.square().sum()).__pow__(0.5) - kruskal_norm(k)) < 1e-8
+    # innerprod vs dense
+    dref = torch.zeros(*small3.dims, dtype=torch.float64)
+    dref.index_put_(tuple(small3.inds), small3.vals.double(), accumulate=True)
+    assert abs(kruskal_innerprod(k, small3) - float((dense * dref).sum())) < 1e-6
