@@ -71,6 +71,8 @@ def main():
     ap.add_argument("--dtype", default="f64", choices=["f64", "f32"])
     ap.add_argument("--device", default="cuda")
     ap.add_argument("--csf", default="all", choices=["one", "two", "all"])
+    ap.add_argument("--profile", action="store_true",
+                    help="print a per-phase breakdown (device-synced timers)")
     ap.add_argument("--gather-tiles", type=int, default=-1,
                     help="dense-tiling buckets for factor-row gathers "
                          "(-1 auto: target <=96MB per phase, 0 off)")
@@ -176,6 +178,15 @@ def main():
     elapsed = time.time() - tic
     if runner is not None:
         runner.finalize(st.norm_x)
+
+    if args.profile:
+        from splatt_amd.utils.timers import TimerRegistry
+        reg = TimerRegistry(sync_device=device.type == "cuda")
+        for _ in range(3):
+            grid_cpd_step(st, it, timers=reg)
+            it += 1
+        if rank == 0:
+            print(reg.report(), file=sys.stderr, flush=True)
 
     # max over ranks
     if world > 1:

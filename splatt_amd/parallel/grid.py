@@ -232,12 +232,17 @@ def grid_cpd_init(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
         norm_x=float(nx.item()))
 
 
-def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True) -> float:
+def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
+                  timers=None) -> float:
     """One ALS iteration. `overlap`: start the mode-m partial-row
     all-reduce asynchronously and form the Gram/Cholesky inverse (which
     do not depend on it) underneath — the comm/compute overlap the
     reference lacks (SURVEY.md §2.4 notes comm is fully synchronous
-    there)."""
+    there). `timers`: an optional TimerRegistry (sync_device=True for
+    honest per-phase times) mirroring the reference's TIMER_MTTKRP /
+    TIMER_MPI_* sections."""
+    from contextlib import nullcontext
+    tm = timers.time if timers is not None else (lambda *_: nullcontext())
     dec = st.dec
     nm = len(dec.global_dims)
     dev = st.buf.device
@@ -246,20 +251,25 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True) -> float:
 
     for m in range(nm):
         mb = st.buf[: dec.chunkn[m]]
-        mttkrp(st.cs, st.factors, m, out=mb)
+        with tm("MTTKRP"):
+            mttkrp(st.cs, st.factors, m, out=mb)
         work = None
         group = dec.layer_groups.get(m)
         if _world() > 1 and group != "solo" and dec.repl(m) > 1:
-            work = dist.all_reduce(mb, group=group, async_op=True)
+            with tm("COMM-POST"):
+                work = dist.all_reduce(mb, group=group, async_op=True)
         # Gram product + inverse are independent of mb: compute under comm
-        G = torch.ones(F, F, dtype=dtype, device=dev)
-        for o in range(nm):
-            if o != m:
-                G *= st.grams[o]
-        Ginv = spd_inverse(G)
+        with tm("SOLVE"):
+            G = torch.ones(F, F, dtype=dtype, device=dev)
+            for o in range(nm):
+                if o != m:
+                    G *= st.grams[o]
+            Ginv = spd_inverse(G)
         if work is not None:
-            work.wait()
-        A = mb @ Ginv
+            with tm("COMM-WAIT"):
+                work.wait()
+        with tm("SOLVE"):
+            A = mb @ Ginv
         # lambda over GLOBAL rows of mode m
         if it == 0:
             s = A.square().sum(dim=0)
