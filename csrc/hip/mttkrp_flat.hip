@@ -446,7 +446,9 @@ mttkrp_flat_generic_kern(const int32_t * __restrict__ key,
 }
 
 inline int64_t pick_span(int64_t nnz) {
-  int64_t span = nnz / 65536;
+  const char * e = getenv("SPLATT_SPAN_WAVES");   // tuning knob
+  const int64_t target_waves = e ? atoll(e) : 65536;
+  int64_t span = nnz / target_waves;
   if (span < 256) span = 256;
   if (span > 16384) span = 16384;
   return span;

@@ -61,7 +61,7 @@ def _stage_blocks(c: Csf) -> dict:
             right=False).cpu()
     else:
         bnd = torch.tensor([0, nnz])
-    tgt = max(4096, nnz // 16384)
+    tgt = max(4096, nnz // int(os.environ.get("SPLATT_LDS_BLOCKS", "16384")))
     starts, ends, row0s = [], [], []
     for b in range(len(bnd) - 1):
         s, e = int(bnd[b]), int(bnd[b + 1])
