@@ -28,7 +28,12 @@ def _gpu_mttkrp_csf(c: Csf, depth: int, mats: List[torch.Tensor],
     nm = c.nmodes
     if nm != 3:
         raise NotImplementedError(
-            f"device MTTKRP currently supports 3-mode CSF, got {nm} modes")
+            f"the hierarchical walk kernels support 3-mode CSF, got {nm} "
+            f"modes (use alg='flat')")
+    if c.fptr[0] is None:
+        raise ValueError(
+            "this CSF was built flat-only (no fptr tree); the hierarchical "
+            "walk kernels need a full build — use alg='flat'")
     perm = c.dim_perm
     which = 0 if depth == 0 else (2 if depth == nm - 1 else 1)
     if which == 0:
