@@ -185,9 +185,12 @@ def _build_csf_device(t: SpTensor, perm: List[int],
             tiles = (dim_big + chunk - 1) // chunk
             # gate: bucketing must not shred the output-key runs, or the
             # atomic-per-run economy is lost (measured: Netflix-shaped
-            # -28% without this check). Require >=32 nnz per (bucket, row).
+            # -28% without this check). Require >= min_run nnz per
+            # (bucket, row); 32 measured as safe default.
+            import os as _os
+            min_run = int(_os.environ.get("SPLATT_STAGE_MIN_RUN", "32"))
             root_dim = max(1, t.dims[perm[0]])
-            if tiles > 1 and nnz // (tiles * root_dim) < 32:
+            if tiles > 1 and nnz // (tiles * root_dim) < min_run:
                 tiles = 1
                 stage_rank = 0
         else:
