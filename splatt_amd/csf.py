@@ -174,27 +174,31 @@ def _build_csf_device(t: SpTensor, perm: List[int],
         order = order.index_select(0, torch.argsort(keys, stable=True))
     stage_meta = None
     if flat_only and (gather_tiles > 1 or stage_rank > 0):
-        # outermost: bucket of the largest non-root mode's row index
         big = max(range(1, nm), key=lambda l: t.dims[perm[l]])
-        dim_big = t.dims[perm[big]]
         if stage_rank > 0:
-            # LDS-staging buckets: rows-per-bucket sized to the LDS budget
-            vbytes = t.vals.element_size()
-            chunk = max(64, (lds_kb * 1024) // (stage_rank * vbytes))
-            chunk = min(chunk, dim_big)
-            tiles = (dim_big + chunk - 1) // chunk
-            # gate: bucketing must not shred the output-key runs, or the
-            # atomic-per-run economy is lost (measured: Netflix-shaped
-            # -28% without this check). Require >= min_run nnz per
-            # (bucket, row); default 8 (measured: Netflix +13% at 8-9 nnz runs,
-            # -28% at ~1 nnz runs).
+            # LDS-staging buckets: rows-per-bucket sized to the LDS budget.
+            # Candidate selection: among non-root levels, stage the LARGEST
+            # dim whose bucketing keeps the output-key runs dense — the
+            # gate (>= min_run nnz per bucket x output row) protects the
+            # one-atomic-per-run economy (measured: Netflix +13% at 8-9
+            # nnz runs, -28% at ~1 nnz runs).
             import os as _os
             min_run = int(_os.environ.get("SPLATT_STAGE_MIN_RUN", "8"))
+            vbytes = t.vals.element_size()
+            chunk_cap = max(64, (lds_kb * 1024) // (stage_rank * vbytes))
             root_dim = max(1, t.dims[perm[0]])
-            if tiles > 1 and nnz // (tiles * root_dim) < min_run:
-                tiles = 1
+            big, chunk, tiles = -1, 0, 1
+            for l in sorted(range(1, nm), key=lambda x: -t.dims[perm[x]]):
+                d = t.dims[perm[l]]
+                ch = min(chunk_cap, d)
+                ti = (d + ch - 1) // ch
+                if ti <= 1 or nnz // (ti * root_dim) >= min_run:
+                    big, chunk, tiles = l, ch, ti
+                    break
+            if big < 0:
                 stage_rank = 0
         else:
+            dim_big = t.dims[perm[big]]
             tiles = gather_tiles
             chunk = (dim_big + tiles - 1) // tiles
         if tiles > 1:
