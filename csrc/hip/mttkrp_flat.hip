@@ -406,15 +406,13 @@ mttkrp_flat4_kern(const int32_t * __restrict__ key,
 // ------------------------------------------------------ generic-rank kernel
 // lane = column (chunked by 64), wave walks its span serially. Correctness
 // path for ranks outside the spec set.
+// up to 7 other modes (8-mode tensors, reference SPLATT_MAX_NMODES);
+// idx/mats arrive as a small device-side pointer table
 template <typename V, int NOTHER>
 __global__ void __launch_bounds__(256)
 mttkrp_flat_generic_kern(const int32_t * __restrict__ key,
-                         const int32_t * __restrict__ i0,
-                         const int32_t * __restrict__ i1,
-                         const int32_t * __restrict__ i2,
-                         const int32_t * __restrict__ i3,
-                         const V * __restrict__ m0, const V * __restrict__ m1,
-                         const V * __restrict__ m2, const V * __restrict__ m3,
+                         const int32_t * const * __restrict__ idx,
+                         const V * const * __restrict__ mats,
                          const V * __restrict__ vals, int64_t nnz,
                          int64_t span, int rank, V * __restrict__ out) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -423,16 +421,20 @@ mttkrp_flat_generic_kern(const int32_t * __restrict__ key,
   const int64_t p0 = wid * span;
   if (p0 >= nnz) return;
   const int64_t p1 = min64(nnz, p0 + span);
+  const int32_t * ip[NOTHER];
+  const V * mp[NOTHER];
+  #pragma unroll
+  for (int t = 0; t < NOTHER; ++t) { ip[t] = idx[t]; mp[t] = mats[t]; }
   for (int cb = 0; cb < rank; cb += WAVE) {
     const int c = cb + lane;
     if (c >= rank) break;
     int32_t cur = key[p0];
     V acc = (V)0;
     for (int64_t p = p0; p < p1; ++p) {
-      V x = vals[p] * m0[(int64_t)i0[p] * rank + c]
-                    * m1[(int64_t)i1[p] * rank + c];
-      if (NOTHER > 2) x *= m2[(int64_t)i2[p] * rank + c];
-      if (NOTHER > 3) x *= m3[(int64_t)i3[p] * rank + c];
+      V x = vals[p];
+      #pragma unroll
+      for (int t = 0; t < NOTHER; ++t)
+        x *= mp[t][(int64_t)ip[t][p] * rank + c];
       const int32_t k = key[p];
       if (k != cur) {
         atomic_add_g(&out[(int64_t)cur * rank + c], acc);
@@ -466,8 +468,8 @@ inline int pick_unroll() {
 }
 
 template <typename V>
-void launch_flat(const int32_t * key, const int32_t * const idx[4],
-                 const V * const mats[4], const V * vals, int64_t nnz,
+void launch_flat(const int32_t * key, const int32_t * const idx[8],
+                 const V * const mats[8], const V * vals, int64_t nnz,
                  V * out, int rank, int nother, hipStream_t st) {
   const int64_t span = pick_span(nnz);
   const int64_t nwaves = (nnz + span - 1) / span;
@@ -496,7 +498,7 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
   switch (rank) { case 4: LU(4, N_); break; case 8: LU(8, N_); break; \
                   case 16: LU(16, N_); break; case 32: LU(32, N_); break; \
                   default: LU(64, N_); break; }
-  if (spec_ok(rank)) {
+  if (spec_ok(rank) && nother <= 4) {
     const int uu = pick_unroll();
     switch (nother) {
       case 2: LF(2); break;
@@ -509,12 +511,25 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
 #undef LU
 #undef L1
 #undef ARGS
-#define GARGS key, idx[0], idx[1], idx[2], idx[3], mats[0], mats[1], mats[2], \
-              mats[3], vals, nnz, span, rank, out
+  // generic path (odd ranks or >5 modes): pointer tables shipped by value
+  // through constant kernel args would need 2x8 slots; simplest is a tiny
+  // device-side table uploaded per call on the same stream.
+  static thread_local void * d_tab = nullptr;
+  if (!d_tab) (void)hipMalloc(&d_tab, 16 * sizeof(void*));
+  void * h_tab[16];
+  for (int t = 0; t < 8; ++t) h_tab[t] = (void*)(t < nother ? idx[t] : nullptr);
+  for (int t = 0; t < 8; ++t) h_tab[8 + t] = (void*)(t < nother ? mats[t] : nullptr);
+  (void)hipMemcpyAsync(d_tab, h_tab, sizeof(h_tab), hipMemcpyHostToDevice, st);
+  const int32_t * const * dixp = (const int32_t * const *)d_tab;
+  const V * const * dmp = (const V * const *)((void**)d_tab + 8);
+#define GARGS key, dixp, dmp, vals, nnz, span, rank, out
   switch (nother) {
     case 2:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 2>), grid, block, 0, st, GARGS); break;
     case 3:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 3>), grid, block, 0, st, GARGS); break;
-    default: hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 4>), grid, block, 0, st, GARGS); break;
+    case 4:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 4>), grid, block, 0, st, GARGS); break;
+    case 5:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 5>), grid, block, 0, st, GARGS); break;
+    case 6:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 6>), grid, block, 0, st, GARGS); break;
+    default: hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 7>), grid, block, 0, st, GARGS); break;
   }
 #undef GARGS
 }
@@ -522,25 +537,19 @@ void launch_flat(const int32_t * key, const int32_t * const idx[4],
 }  // namespace
 
 extern "C" void splatt_hip_mttkrp_flat_f64(
-    const int32_t * key, const int32_t * i0, const int32_t * i1,
-    const int32_t * i2, const int32_t * i3, const double * m0,
-    const double * m1, const double * m2, const double * m3,
+    const int32_t * key, const int32_t * const * idx,
+    const double * const * mats,
     const double * vals, int64_t nnz, double * out, int rank, int nother,
     void * stream) {
-  const int32_t * idx[4] = {i0, i1, i2, i3};
-  const double * mats[4] = {m0, m1, m2, m3};
   launch_flat<double>(key, idx, mats, vals, nnz, out, rank, nother,
                       (hipStream_t)stream);
 }
 
 extern "C" void splatt_hip_mttkrp_flat_f32(
-    const int32_t * key, const int32_t * i0, const int32_t * i1,
-    const int32_t * i2, const int32_t * i3, const float * m0,
-    const float * m1, const float * m2, const float * m3,
+    const int32_t * key, const int32_t * const * idx,
+    const float * const * mats,
     const float * vals, int64_t nnz, float * out, int rank, int nother,
     void * stream) {
-  const int32_t * idx[4] = {i0, i1, i2, i3};
-  const float * mats[4] = {m0, m1, m2, m3};
   launch_flat<float>(key, idx, mats, vals, nnz, out, rank, nother,
                      (hipStream_t)stream);
 }

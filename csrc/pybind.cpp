@@ -386,13 +386,11 @@ void splatt_hip_mttkrp_leaf3_f32(
 int splatt_hip_kernels_arch(void);
 // flat (expanded-CSF) kernels, csrc/hip/mttkrp_flat.hip
 void splatt_hip_mttkrp_flat_f64(
-    const int32_t*, const int32_t*, const int32_t*, const int32_t*,
-    const int32_t*, const double*, const double*, const double*,
-    const double*, const double*, int64_t, double*, int, int, void*);
+    const int32_t*, const int32_t* const*, const double* const*,
+    const double*, int64_t, double*, int, int, void*);
 void splatt_hip_mttkrp_flat_f32(
-    const int32_t*, const int32_t*, const int32_t*, const int32_t*,
-    const int32_t*, const float*, const float*, const float*,
-    const float*, const float*, int64_t, float*, int, int, void*);
+    const int32_t*, const int32_t* const*, const float* const*,
+    const float*, int64_t, float*, int, int, void*);
 // dense kernels, csrc/hip/dense_kernels.hip
 void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
 void splatt_hip_gram_f32(const float*, int64_t, int, float*, void*);
@@ -468,25 +466,23 @@ static void py_gpu_mttkrp_flat(Tensor key, std::vector<Tensor> idx,
                                std::vector<Tensor> mats, Tensor vals,
                                Tensor out, int64_t stream) {
   const int nother = (int)idx.size();
-  TORCH_CHECK(nother >= 2 && nother <= 4, "flat kernel supports 3..5 modes");
+  TORCH_CHECK(nother >= 2 && nother <= 7, "flat kernel supports 3..8 modes");
   TORCH_CHECK((int)mats.size() == nother);
   const int rank = (int)mats[0].size(1);
   const int64_t nnz = vals.numel();
-  const int32_t * ip[4] = {nullptr, nullptr, nullptr, nullptr};
+  const int32_t * ip[8] = {};
   for (int t = 0; t < nother; ++t) ip[t] = idx[t].data_ptr<int32_t>();
   if (vals.scalar_type() == torch::kFloat64) {
-    const double * mp[4] = {nullptr, nullptr, nullptr, nullptr};
+    const double * mp[8] = {};
     for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
-    splatt_hip_mttkrp_flat_f64(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
-                               ip[3], mp[0], mp[1], mp[2], mp[3],
+    splatt_hip_mttkrp_flat_f64(key.data_ptr<int32_t>(), ip, mp,
                                vals.data_ptr<double>(), nnz,
                                out.data_ptr<double>(), rank, nother,
                                (void*)stream);
   } else {
-    const float * mp[4] = {nullptr, nullptr, nullptr, nullptr};
+    const float * mp[8] = {};
     for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
-    splatt_hip_mttkrp_flat_f32(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
-                               ip[3], mp[0], mp[1], mp[2], mp[3],
+    splatt_hip_mttkrp_flat_f32(key.data_ptr<int32_t>(), ip, mp,
                                vals.data_ptr<float>(), nnz,
                                out.data_ptr<float>(), rank, nother,
                                (void*)stream);

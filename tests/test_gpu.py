@@ -203,3 +203,16 @@ def test_gpu_graph_step_matches_eager(t3):
     for it in range(6):  # 1 eager + 2 warmup-in-capture + 3 replays
         fit_eager = grid_cpd_step(st2, it)
     assert abs(fit_graph - fit_eager) < 1e-6, (fit_graph, fit_eager)
+
+
+def test_gpu_mttkrp_flat_8mode():
+    """8-mode tensors (SPLATT_MAX_NMODES) on the generic device path."""
+    dims = [8, 9, 10, 7, 6, 8, 5, 9]
+    t = sp.SpTensor.synthetic(dims, 30_000, seed=77)
+    mats_c = make_mats(t.dims, 6)
+    mats_g = [m.cuda() for m in mats_c]
+    cs = sp.csf_alloc(t.to("cuda"), "one")
+    for mode in (0, 3, 7):
+        out = sp.mttkrp(cs, mats_g, mode)
+        ref = sp.mttkrp_stream(t, mats_c, mode)
+        assert (out.cpu() - ref).abs().max().item() < 1e-8, mode
