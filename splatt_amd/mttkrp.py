@@ -98,7 +98,8 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
     key = c.ancestor_expand(depth)
     rank = int(mats[0].shape[1])
     stream = torch.cuda.current_stream().cuda_stream
-    use_lds = (depth == 0 and getattr(c, "_stage", None) is not None
+    use_lds = (depth == 0 and nm <= 5
+               and getattr(c, "_stage", None) is not None
                and rank in (4, 8, 16, 32, 64)
                and os.environ.get("SPLATT_NO_LDS") != "1")
     if use_lds:
