@@ -207,3 +207,44 @@ def test_load_shard_and_write_factors(tmp_path):
     for m in range(3):
         assert (tmp_path / f"mode{m + 1}.mat").exists()
     assert (tmp_path / "lambda.mat").exists()
+
+
+def _w8_worker(rank, world, file_store, result_q, grid):
+    torch.distributed.init_process_group(
+        "gloo", init_method=f"file://{file_store}", rank=rank,
+        world_size=world)
+    try:
+        from splatt_amd.parallel.grid import GridDecomp, grid_cpd_als
+        from splatt_amd.parallel.dist_cpd import build_shard_csf
+        t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+        dec = GridDecomp.create(list(DIMS), grid=grid)
+        shard = dec.localize(t)
+        cs = build_shard_csf(shard, list(DIMS), "two")
+        opts = sp.CpdOptions(max_iters=3, tolerance=0.0, seed=SEED)
+        k = grid_cpd_als(cs, dec, RANK_F, opts)
+        if rank == 0:
+            result_q.put(k.fit)
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("grid", [[1, 1, 8], [2, 1, 4]])
+def test_world8_matches_single(tmp_path, grid):
+    """The round-end 8-rank topologies (coarse + medium), over gloo."""
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+    k1 = sp.cpd_als(t, RANK_F, sp.CpdOptions(max_iters=3, tolerance=0.0,
+                                             seed=SEED))
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    store = str(tmp_path / f"store8_{grid[0]}_{grid[2]}")
+    procs = [ctx.Process(target=_w8_worker, args=(r, 8, store, q, grid))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    fit8 = q.get()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    assert abs(fit8 - k1.fit) < 1e-8, (fit8, k1.fit, grid)
