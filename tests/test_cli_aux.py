@@ -148,3 +148,25 @@ def test_legacy_bench_algs():
                        validate=True)
     for alg in ("csf", "giga", "ttbox"):
         assert res[alg]["validated"], alg
+
+
+def test_bench_contract_json(tmp_path):
+    """bench.py must emit exactly one valid JSON line with the driver's
+    contract fields (CPU small config)."""
+    import json
+    import subprocess
+    import sys as _sys
+    r = subprocess.run(
+        [_sys.executable, "bench.py", "--device", "cpu", "--config", "small",
+         "--steps", "1", "--warmup", "1"],
+        capture_output=True, text=True, cwd=ROOT, timeout=300)
+    assert r.returncode == 0, r.stderr[-500:]
+    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    j = json.loads(lines[0])
+    for field in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                  "ms_per_step", "higher_is_better", "scaling",
+                  "vs_baseline", "dtype", "data", "config"):
+        assert field in j, field
+    assert j["n_gpus"] == 1 and j["steps"] == 1
+    assert j["higher_is_better"] is True
