@@ -154,12 +154,14 @@ def test_bench_contract_json(tmp_path):
     """bench.py must emit exactly one valid JSON line with the driver's
     contract fields (CPU small config)."""
     import json
+    import os
     import subprocess
     import sys as _sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     r = subprocess.run(
         [_sys.executable, "bench.py", "--device", "cpu", "--config", "small",
          "--steps", "1", "--warmup", "1"],
-        capture_output=True, text=True, cwd=ROOT, timeout=300)
+        capture_output=True, text=True, cwd=root, timeout=300)
     assert r.returncode == 0, r.stderr[-500:]
     lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
     assert len(lines) == 1
