@@ -44,10 +44,13 @@ CONFIGS = {
 }
 
 
-def synth_box_shard(dec, nnz, seed, dtype, device):
+def synth_box_shard(dec, nnz, seed, dtype, device, dist="uniform"):
     """Generate this rank's grid-box shard directly at local size (indices
     already chunk-local)."""
     local_dims = list(dec.chunkn)
+    if dist == "zipf":
+        return sp.SpTensor.synthetic(local_dims, nnz, dtype=dtype,
+                                     device=device, seed=seed, dist="zipf")
     gen_dev = device if nnz > 200_000_000 else "cpu"
     g = torch.Generator(device=gen_dev).manual_seed(seed)
     cols = []
@@ -71,6 +74,8 @@ def main():
     ap.add_argument("--dtype", default="f64", choices=["f64", "f32"])
     ap.add_argument("--device", default="cuda")
     ap.add_argument("--csf", default="all", choices=["one", "two", "all"])
+    ap.add_argument("--synth", default="uniform", choices=["uniform", "zipf"],
+                    help="index distribution of the synthetic tensor")
     ap.add_argument("--profile", action="store_true",
                     help="print a per-phase breakdown (device-synced timers)")
     ap.add_argument("--gather-tiles", type=int, default=-1,
@@ -115,7 +120,7 @@ def main():
         scaling = "strong"
     dec = GridDecomp.create(global_dims, grid=grid)
     shard = synth_box_shard(dec, nnz_local, seed=0xB0B0 + rank, dtype=dtype,
-                            device=device)
+                            device=device, dist=args.synth)
     gt = max(args.gather_tiles, 0)
     stage_rank = rank_f if (args.gather_tiles < 0
                             and device.type == "cuda") else 0

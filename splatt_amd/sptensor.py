@@ -88,21 +88,37 @@ class SpTensor:
                   dtype: torch.dtype = torch.float64,
                   device: str | torch.device = "cpu",
                   seed: int = 0x5eed,
-                  concentration: float = 1.0) -> "SpTensor":
+                  concentration: float = 1.0,
+                  dist: str = "uniform",
+                  zipf_a: float = 1.1) -> "SpTensor":
         """Random synthetic tensor of a target shape/nnz.
 
-        `concentration` > 1 skews index draws toward low ids (a crude
-        power-law-ish density profile so fibers have non-uniform lengths,
-        which is the load-balance regime MTTKRP scheduling must handle).
-        Duplicates are NOT merged (matches how nnz counts are quoted).
+        `dist`: "uniform" draws each index independently (hardest case
+        for CSF compression — near-singleton fibers); "zipf" draws each
+        mode's indices from a power law with exponent `zipf_a`, giving
+        the heavy slices / long fibers real tensors show (the
+        load-balance regime the reference's CCP + privatization target).
+        `concentration` > 1 additionally skews uniform draws toward low
+        ids. Duplicates are NOT merged (matches how nnz counts are
+        quoted).
         """
         g = torch.Generator(device="cpu").manual_seed(seed)
         cols = []
         for m, d in enumerate(dims):
             u = torch.rand(nnz, generator=g, dtype=torch.float64)
-            if concentration != 1.0:
-                u = u.pow(concentration)
-            cols.append((u * d).long().clamp_(0, d - 1))
+            if dist == "zipf":
+                # inverse-CDF of a bounded Pareto on [1, d]
+                a = zipf_a
+                lo, hi = 1.0, float(d)
+                x = (hi ** (1 - a) - lo ** (1 - a)) * u + lo ** (1 - a)
+                idx = x.pow(1.0 / (1 - a)).long() - 1
+                # random relabeling so heavy slices are spread over ids
+                perm = torch.randperm(d, generator=g)
+                cols.append(perm[idx.clamp_(0, d - 1)])
+            else:
+                if concentration != 1.0:
+                    u = u.pow(concentration)
+                cols.append((u * d).long().clamp_(0, d - 1))
         inds = torch.stack(cols, 0)
         vals = torch.rand(nnz, generator=g, dtype=torch.float64).to(dtype)
         t = SpTensor(inds, vals, list(dims))
