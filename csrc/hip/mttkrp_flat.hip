@@ -519,7 +519,10 @@ void launch_flat(const int32_t * key, const int32_t * const idx[8],
   void * h_tab[16];
   for (int t = 0; t < 8; ++t) h_tab[t] = (void*)(t < nother ? idx[t] : nullptr);
   for (int t = 0; t < 8; ++t) h_tab[8 + t] = (void*)(t < nother ? mats[t] : nullptr);
-  (void)hipMemcpyAsync(d_tab, h_tab, sizeof(h_tab), hipMemcpyHostToDevice, st);
+  // the prior user of d_tab may still be queued on `st`: order the update
+  // behind it, and use a blocking copy (pageable h_tab is a stack array)
+  (void)hipStreamSynchronize(st);
+  (void)hipMemcpy(d_tab, h_tab, sizeof(h_tab), hipMemcpyHostToDevice);
   const int32_t * const * dixp = (const int32_t * const *)d_tab;
   const V * const * dmp = (const V * const *)((void**)d_tab + 8);
 #define GARGS key, dixp, dmp, vals, nnz, span, rank, out
