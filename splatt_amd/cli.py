@@ -29,6 +29,10 @@ def _load(args) -> "sp.SpTensor":
 
 
 def cmd_cpd(args) -> int:
+    import os
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        return _cmd_cpd_dist(args, world)
     t = _load(args)
     print(stats_tt(t, args.tensor))
     opts = sp.CpdOptions(tolerance=args.tol, max_iters=args.its,
@@ -55,6 +59,41 @@ def cmd_cpd(args) -> int:
             for x in k.lam.cpu().tolist():
                 fh.write(f"{x:.17g}\n")
     print(TIMERS.report())
+    return 0
+
+
+def _cmd_cpd_dist(args, world: int) -> int:
+    """`torchrun --nproc-per-node N -m splatt_amd cpd ...` — the analog of
+    `mpirun splatt cpd` (reference cmds/mpi_cmd_cpd.c:175): grid
+    decomposition over RCCL (GPU) or gloo (CPU), rank-0 output."""
+    import os
+    import torch.distributed as dist
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    from splatt_amd.parallel.grid import (GridDecomp, grid_cpd_als,
+                                          load_shard, write_factors)
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = args.device in ("auto", "cuda") and torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
+    dist.init_process_group("nccl" if use_cuda else "gloo")
+    dtype = torch.float64 if args.dtype == "f64" else torch.float32
+    t = sp.SpTensor.load(args.tensor, dtype)
+    dec = GridDecomp.create(list(t.dims))
+    shard = dec.localize(t)
+    if use_cuda:
+        shard = shard.to(f"cuda:{local_rank % torch.cuda.device_count()}")
+    cs = build_shard_csf(shard, list(t.dims), args.csf,
+                         flat_only=use_cuda)
+    opts = sp.CpdOptions(tolerance=args.tol, max_iters=args.its,
+                         seed=args.seed, csf_alloc=args.csf)
+    k = grid_cpd_als(cs, dec, args.rank, opts)
+    if rank == 0:
+        print(f"Final fit: {k.fit:.5f}  (iterations: {k.niters}; "
+              f"grid {dec.grid})")
+    if not args.nowrite:
+        write_factors(k, dec)
+    dist.destroy_process_group()
     return 0
 
 
