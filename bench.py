@@ -91,6 +91,11 @@ def main():
         rank_f = args.rank_f
     dtype = torch.float64 if args.dtype == "f64" else torch.float32
 
+    if args.device != "cpu" and not torch.cuda.is_available():
+        print("bench.py: no GPU visible; pass --device cpu for a host run",
+              file=sys.stderr)
+        return 2
+
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
