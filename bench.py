@@ -238,4 +238,4 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    sys.exit(main())
