@@ -51,7 +51,7 @@ mttkrp_flat5_kern(const int32_t * __restrict__ key,
                   const int32_t * __restrict__ blk_row0,
                   int32_t chunk, int32_t dim0,
                   V * __restrict__ out) {
-  constexpr int GB = 8;
+  constexpr int GB = 6;   // 77 VGPR -> 6 waves/SIMD; LDS then caps at 24/CU
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   V * smem = reinterpret_cast<V *>(smem_raw);
 
