@@ -313,7 +313,8 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
 def comm_stats(dec: GridDecomp, shard_nnz: int, rank_f: int,
                val_bytes: int = 8) -> dict:
     """Per-rank communication volume per ALS iteration + nnz balance
-    (reference mpi_rank_stats / mpi_cpd_stats, stats.c:298-465)."""
+    (reference mpi_rank_stats / mpi_cpd_stats, stats.c:298-465).
+    COLLECTIVE: every rank must call this (it all-reduces nnz counts)."""
     world = _world()
     per_mode = []
     total = 0
