@@ -142,7 +142,20 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
         c = src
         depth = c.level_of_mode(mode)
 
+    if not 0 <= mode < c.nmodes:
+        raise IndexError(f"mode {mode} out of range for {c.nmodes} modes")
+    if len(mats) != c.nmodes:
+        raise ValueError(f"need {c.nmodes} factor matrices, got {len(mats)}")
     rank = int(mats[0].shape[1])
+    for m, A in enumerate(mats):
+        if A.shape != (c.dims[m], rank):
+            raise ValueError(
+                f"factor {m} has shape {tuple(A.shape)}, expected "
+                f"({c.dims[m]}, {rank})")
+        if A.dtype != c.vals.dtype or A.device != c.device:
+            raise ValueError(
+                f"factor {m}: dtype/device {A.dtype}/{A.device} does not "
+                f"match tensor {c.vals.dtype}/{c.device}")
     if out is None:
         out = torch.empty(c.dims[mode], rank, dtype=mats[0].dtype,
                           device=mats[0].device)
