@@ -57,12 +57,22 @@ class Csf:
         return int(fp.numel() - 1) if fp is not None else 0
 
     def to(self, device) -> "Csf":
-        return Csf(
+        c = Csf(
             list(self.dims), list(self.dim_perm),
             [t.to(device) if t is not None else None for t in self.fptr],
             [t.to(device) if t is not None else None for t in self.fids],
             self.vals.to(device),
         )
+        # carry expansions + stage metadata (a frozen flat CSF has no fptr
+        # tree to recompute them from)
+        cache = getattr(self, "_expand_cache", None)
+        if cache:
+            object.__setattr__(c, "_expand_cache",
+                               {l: t.to(device) for l, t in cache.items()})
+        stage = getattr(self, "_stage", None)
+        if stage is not None:
+            object.__setattr__(c, "_stage", dict(stage))
+        return c
 
     def storage_bytes(self) -> int:
         b = self.vals.numel() * self.vals.element_size()

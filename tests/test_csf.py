@@ -89,3 +89,13 @@ def test_freeze_flat_keeps_expansions(small3):
     assert all(fp is None for fp in c.fptr)
     for l in range(3):
         assert torch.equal(c.ancestor_expand(l), want[l])
+
+
+def test_frozen_csf_survives_device_move(small3):
+    """to(device) must carry expansions + stage metadata for tree-less
+    (frozen) CSF builds."""
+    perm = sp.order_modes(small3.dims, "smallfirst")
+    c = build_csf(small3, perm).freeze_flat()
+    c2 = c.to("cpu")
+    for l in range(3):
+        assert torch.equal(c.ancestor_expand(l), c2.ancestor_expand(l))
