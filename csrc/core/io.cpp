@@ -147,6 +147,8 @@ SpTensor<V> bin_read(const std::string & path) {
       fread(&nm, 8, 1, f.get()) != 1 || fread(&nnz, 8, 1, f.get()) != 1)
     throw std::runtime_error("truncated header in " + path);
   if (nm < 1 || nm > MAX_NMODES) throw std::runtime_error("bad nmodes");
+  if ((ib != 4 && ib != 8) || (vb != 4 && vb != 8))
+    throw std::runtime_error("bad index/value width in binary header");
   idx_t dims[MAX_NMODES];
   for (uint64_t m = 0; m < nm; ++m) {
     uint64_t d;
