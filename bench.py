@@ -79,14 +79,15 @@ def main():
     ap.add_argument("--profile", action="store_true",
                     help="print a per-phase breakdown (device-synced timers)")
     ap.add_argument("--gather-tiles", type=int, default=-1,
-                    help="dense-tiling buckets for factor-row gathers "
-                         "(-1 auto: target <=96MB per phase, 0 off)")
+                    help="-1 (default): LDS-staged bucketing sized by "
+                         "SPLATT_LDS_KB; 0: off; N>1: plain gather-range "
+                         "buckets without LDS staging")
     ap.add_argument("--decomp", default="coarse", choices=["coarse", "medium"],
                     help="coarse = 1D layers on the longest mode (weak "
                          "scaling); medium = nmodes-D grid (strong scaling)")
     args = ap.parse_args()
 
-    dims, nnz_shard, rank_f, conc = CONFIGS[args.config]
+    dims, nnz_shard, rank_f, _ = CONFIGS[args.config]
     if args.rank_f:
         rank_f = args.rank_f
     dtype = torch.float64 if args.dtype == "f64" else torch.float32
