@@ -108,3 +108,24 @@ def test_cli_binary(tmp_path, small3):
     r = subprocess.run([EXE, "stats", str(tmp_path / "t.bin")],
                        capture_output=True, text=True)
     assert r.returncode == 0 and "NNZ" in r.stdout
+
+
+def test_csf_convert_in_memory(lib, small3):
+    """splatt_csf_convert from raw index/value arrays."""
+    import numpy as np
+    nnz = small3.nnz
+    inds = [small3.inds[m].numpy().astype(np.uint64).copy() for m in range(3)]
+    vals = small3.vals.double().numpy().copy()
+    arr_t = ctypes.POINTER(ctypes.c_uint64) * 3
+    ptrs = arr_t(*[i.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
+                   for i in inds])
+    o = lib.splatt_default_opts()
+    csf = ctypes.c_void_p()
+    rc = lib.splatt_csf_convert(3, nnz, ptrs,
+                                vals.ctypes.data_as(
+                                    ctypes.POINTER(ctypes.c_double)),
+                                ctypes.byref(csf), o)
+    assert rc == 0
+    assert lib.splatt_csf_nnz(csf) == nnz
+    lib.splatt_free_csf(csf, o)
+    lib.splatt_free_opts(o)

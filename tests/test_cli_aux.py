@@ -196,3 +196,16 @@ def test_cli_reorder_partfile(tmp_path, small3):
     part2.write_text("\n".join(str(i % 3) for i in range(small3.nnz)))
     assert cli_main(["reorder", tns, str(tmp_path / "r2.tns"),
                      "--type", "hgraph", "--partfile", str(part2)]) == 0
+
+
+def test_hgraph_fib_and_perm_matrix(small3):
+    from splatt_amd.graph import hgraph_fib
+    hg = hgraph_fib(small3, mode=2)
+    assert hg.nvtxs > 0
+    assert int(hg.eptr[-1]) == hg.eind.numel()
+    # every pin is a valid fiber id
+    assert int(hg.eind.max()) < hg.nvtxs
+    perm = ro.perm_rand(small3.dims, seed=1)
+    A = torch.rand(small3.dims[0], 4, dtype=torch.float64)
+    B = ro.perm_matrix(A, perm.perms[0])
+    assert torch.equal(B[perm.iperms[0]], A)

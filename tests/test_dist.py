@@ -104,12 +104,13 @@ def test_grid_localize():
     assert total == t.nnz
 
 
-def _grid_worker(rank, world, file_store, result_q):
+def _grid_worker(rank, world, file_store, result_q, out_prefix=None):
     torch.distributed.init_process_group(
         "gloo", init_method=f"file://{file_store}", rank=rank,
         world_size=world)
     try:
-        from splatt_amd.parallel.grid import GridDecomp, grid_cpd_als
+        from splatt_amd.parallel.grid import (GridDecomp, grid_cpd_als,
+                                              write_factors)
         from splatt_amd.parallel.dist_cpd import build_shard_csf
         t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
         dec = GridDecomp.create(list(DIMS), grid=[2, 1, 2])
@@ -117,6 +118,8 @@ def _grid_worker(rank, world, file_store, result_q):
         cs = build_shard_csf(shard, list(DIMS), "two")
         opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
         k = grid_cpd_als(cs, dec, RANK_F, opts)
+        if out_prefix:
+            write_factors(k, dec, prefix=out_prefix)
         if rank == 0:
             result_q.put(("fit", k.fit, k.niters))
     finally:
@@ -133,7 +136,8 @@ def test_grid_cpd_matches_single_process(tmp_path):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     store = str(tmp_path / "store_grid")
-    procs = [ctx.Process(target=_grid_worker, args=(r, 4, store, q))
+    prefix = str(tmp_path / "gout_")
+    procs = [ctx.Process(target=_grid_worker, args=(r, 4, store, q, prefix))
              for r in range(4)]
     for p in procs:
         p.start()
@@ -143,6 +147,10 @@ def test_grid_cpd_matches_single_process(tmp_path):
         assert p.exitcode == 0
     assert niters4 == k1.niters
     assert abs(fit4 - k1.fit) < 1e-8, (fit4, k1.fit)
+    # rank-0 wrote GLOBAL-row factor files (chunked modes gathered)
+    for m, d in enumerate(DIMS):
+        lines = open(f"{prefix}mode{m + 1}.mat").read().strip().splitlines()
+        assert len(lines) == d, (m, len(lines))
 
 
 def _fine_worker(rank, world, file_store, result_q):
