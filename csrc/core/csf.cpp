@@ -88,6 +88,10 @@ template <typename V>
 Csf<V> csf_build(SpTensor<V> & tt, const int * perm) {
   Csf<V> c;
   const int nm = tt.nmodes;
+  for (int m = 0; m < nm; ++m)
+    if (tt.dims[m] > 0xFFFFFFFFull)
+      throw std::runtime_error(
+          "mode dimension above 2^32 (32-bit CSF node ids)");
   c.nmodes = nm;
   c.nnz = tt.nnz;
   for (int m = 0; m < nm; ++m) c.dims[m] = tt.dims[m];

@@ -176,6 +176,7 @@ def main(argv=None) -> int:
         prog="splatt",
         description="MI355X-native sparse tensor factorization "
                     "(CPD-ALS / CSF MTTKRP)")
+    ap.add_argument("--version", action="version", version=sp.__version__)
     sub = ap.add_subparsers(dest="cmd", required=True)
 
     p = sub.add_parser("cpd", help="compute the CPD of a sparse tensor")

@@ -147,6 +147,9 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
     (reference tt_densetile's cache story, tile.c:262, recast for the
     per-XCD L2/L3 hierarchy). Output-key runs stay contiguous per bucket,
     so the kernel is unchanged (one atomic per key run per bucket)."""
+    if any(d > 0xFFFFFFFF for d in t.dims):
+        raise ValueError("mode dimensions above 2^32 are not supported "
+                         "(CSF node ids are 32-bit; shard the mode first)")
     if t.device.type == "cuda":
         if lds_kb <= 0:
             import os

@@ -60,3 +60,34 @@ def test_fixed_dedup_compress():
     assert sorted(f.vals.tolist()) == [3.0, 7.0]
     assert f.dims == [2, 2, 2]  # empty slices removed
     assert f.indmaps[0].tolist() == [1, 5]
+
+
+def test_corrupt_binary_rejected(tmp_path):
+    """Truncated/garbage binary files raise clean errors, never crash."""
+    import pytest
+    p = tmp_path / "bad.bin"
+    p.write_bytes(b"SPLATTB1" + b"\xff" * 16)   # bad widths
+    with pytest.raises(RuntimeError):
+        sp.load(str(p))
+    p2 = tmp_path / "bad2.bin"
+    p2.write_bytes(b"NOTMAGIC" + b"\x00" * 64)
+    with pytest.raises(RuntimeError):
+        sp.load(str(p2))
+    p3 = tmp_path / "trunc.bin"
+    t = sp.SpTensor.synthetic([5, 5, 5], 50, seed=1)
+    t.save(str(p3))
+    data = p3.read_bytes()
+    p3.write_bytes(data[: len(data) // 2])
+    with pytest.raises(RuntimeError):
+        sp.load(str(p3))
+
+
+def test_huge_dim_rejected():
+    import pytest
+    from splatt_amd.csf import build_csf
+    import torch
+    t = sp.SpTensor(torch.zeros(2, 1, dtype=torch.int64),
+                    torch.ones(1, dtype=torch.float64),
+                    [2, 2**33])
+    with pytest.raises((ValueError, RuntimeError)):
+        build_csf(t, [0, 1])
