@@ -200,9 +200,15 @@ def _build_csf_device(t: SpTensor, perm: List[int],
             vbytes = t.vals.element_size()
             chunk_cap = max(64, (lds_kb * 1024) // (stage_rank * vbytes))
             root_dim = max(1, t.dims[perm[0]])
+            # staging a factor that already fits the per-XCD L2 buys no
+            # gather relief but costs LDS occupancy: require >= min_bytes
+            min_bytes = int(_os.environ.get("SPLATT_STAGE_MIN_BYTES",
+                                            str(2 * 1024 * 1024)))
             big, chunk, tiles = -1, 0, 1
             for l in sorted(range(1, nm), key=lambda x: -t.dims[perm[x]]):
                 d = t.dims[perm[l]]
+                if d * stage_rank * vbytes < min_bytes:
+                    continue
                 ch = min(chunk_cap, d)
                 ti = (d + ch - 1) // ch
                 if ti <= 1 or nnz // (ti * root_dim) >= min_run:
