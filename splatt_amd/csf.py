@@ -200,10 +200,10 @@ def _build_csf_device(t: SpTensor, perm: List[int],
             vbytes = t.vals.element_size()
             chunk_cap = max(64, (lds_kb * 1024) // (stage_rank * vbytes))
             root_dim = max(1, t.dims[perm[0]])
-            # staging a factor that already fits the per-XCD L2 buys no
-            # gather relief but costs LDS occupancy: require >= min_bytes
-            min_bytes = int(_os.environ.get("SPLATT_STAGE_MIN_BYTES",
-                                            str(2 * 1024 * 1024)))
+            # note: even L2-resident factors profit from staging (the TA
+            # tag path, not bandwidth, is the binder — profiles/); the
+            # size gate is available for experiments but off by default
+            min_bytes = int(_os.environ.get("SPLATT_STAGE_MIN_BYTES", "0"))
             big, chunk, tiles = -1, 0, 1
             for l in sorted(range(1, nm), key=lambda x: -t.dims[perm[x]]):
                 d = t.dims[perm[l]]
