@@ -56,6 +56,81 @@ gram_kern(const V * __restrict__ A, int64_t n, int F, int64_t rows_per_blk,
   }
 }
 
+// ------------------------------------------------------------ MFMA gram
+// G = A^T A on the matrix cores (v_mfma_f64_16x16x4_f64 /
+// v_mfma_f32_16x16x4_f32). Key trick: for these shapes the A-operand
+// layout is A[m=lane&15][k=lane>>4] and the B-operand layout is
+// B[k=lane>>4][n=lane&15] — mutual transposes — so feeding the SAME
+// per-lane value x = chunk[k][col] as both operands computes
+// chunk^T * chunk directly. Each wave folds its row range 4 rows per
+// MFMA into (F/16)^2 16x16 accumulator tiles (upper-triangular block
+// pairs only; the mirror is written in the epilogue).
+template <typename V> struct MfmaAcc;
+template <> struct MfmaAcc<double> {
+  using type = __attribute__((ext_vector_type(4))) double;
+  static __device__ __forceinline__ type mfma(double a, double b, type c) {
+    return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct MfmaAcc<float> {
+  using type = __attribute__((ext_vector_type(4))) float;
+  static __device__ __forceinline__ type mfma(float a, float b, type c) {
+    return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+};
+
+template <typename V, int F>
+__global__ void __launch_bounds__(256)
+gram_mfma_kern(const V * __restrict__ A, int64_t n, int64_t rows_per_blk,
+               V * __restrict__ G) {
+  constexpr int NB = F / 16;              // 16-wide column blocks
+  constexpr int NP = NB * (NB + 1) / 2;   // upper-triangular block pairs
+  using Acc = typename MfmaAcc<V>::type;
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int c = lane & 15;                // column within a 16-block
+  const int k = lane >> 4;                // row within the K=4 chunk
+  const int64_t r0 = (int64_t)blockIdx.x * rows_per_blk + (int64_t)wv * 4;
+  const int64_t r1 = min((int64_t)blockIdx.x * rows_per_blk + rows_per_blk, n);
+
+  Acc acc[NP];
+  #pragma unroll
+  for (int p = 0; p < NP; ++p) acc[p] = Acc{};
+
+  for (int64_t r = r0; r < r1; r += 4 * 4 /* waves */) {
+    V v[NB];
+    const int64_t row = r + k;
+    #pragma unroll
+    for (int b = 0; b < NB; ++b)
+      v[b] = row < r1 ? A[row * F + b * 16 + c] : (V)0;
+    int p = 0;
+    #pragma unroll
+    for (int bi = 0; bi < NB; ++bi) {
+      #pragma unroll
+      for (int bj = bi; bj < NB; ++bj) {
+        acc[p] = MfmaAcc<V>::mfma(v[bi], v[bj], acc[p]);
+        ++p;
+      }
+    }
+  }
+
+  // C/D layout of the 16x16x4 forms: reg i -> row (lane>>4)*4+i, col lane&15
+  int p = 0;
+  #pragma unroll
+  for (int bi = 0; bi < NB; ++bi) {
+    #pragma unroll
+    for (int bj = bi; bj < NB; ++bj, ++p) {
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = bi * 16 + (lane >> 4) * 4 + i;
+        const int col = bj * 16 + c;
+        atomic_add_g(&G[row * F + col], acc[p][i]);
+        if (bi != bj) atomic_add_g(&G[col * F + row], acc[p][i]);
+      }
+    }
+  }
+}
+
 template <typename V>
 void launch_gram(const V * A, int64_t n, int F, V * G, hipStream_t st) {
   // enough blocks to fill the chip, >= 4 LDS tiles of work each
@@ -64,6 +139,14 @@ void launch_gram(const V * A, int64_t n, int F, V * G, hipStream_t st) {
   if (nblocks < 1) nblocks = 1;
   const int64_t rows_per_blk = (n + nblocks - 1) / nblocks;
   dim3 grid((uint32_t)nblocks), block(TPB);
+  // matrix-core path for the MFMA-tileable ranks
+  if (F == 16 || F == 32 || F == 64) {
+    switch (F) {
+      case 16: hipLaunchKernelGGL((gram_mfma_kern<V, 16>), grid, block, 0, st, A, n, rows_per_blk, G); return;
+      case 32: hipLaunchKernelGGL((gram_mfma_kern<V, 32>), grid, block, 0, st, A, n, rows_per_blk, G); return;
+      default: hipLaunchKernelGGL((gram_mfma_kern<V, 64>), grid, block, 0, st, A, n, rows_per_blk, G); return;
+    }
+  }
   const int ff = F * F;
   if (ff <= 256)
     hipLaunchKernelGGL((gram_kern<V, 1>), grid, block, 0, st, A, n, F, rows_per_blk, G);
