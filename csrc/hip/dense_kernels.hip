@@ -114,7 +114,8 @@ gram_mfma_kern(const V * __restrict__ A, int64_t n, int64_t rows_per_blk,
     }
   }
 
-  // C/D layout of the 16x16x4 forms: reg i -> row (lane>>4)*4+i, col lane&15
+  // C/D layout of the 16x16x4 forms (probed on gfx950, profiles/):
+  // reg i -> row (lane>>4) + 4*i, col lane&15
   int p = 0;
   #pragma unroll
   for (int bi = 0; bi < NB; ++bi) {
@@ -122,7 +123,7 @@ gram_mfma_kern(const V * __restrict__ A, int64_t n, int64_t rows_per_blk,
     for (int bj = bi; bj < NB; ++bj, ++p) {
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int row = bi * 16 + (lane >> 4) * 4 + i;
+        const int row = bi * 16 + (lane >> 4) + 4 * i;
         const int col = bj * 16 + c;
         atomic_add_g(&G[row * F + col], acc[p][i]);
         if (bi != bj) atomic_add_g(&G[col * F + row], acc[p][i]);
