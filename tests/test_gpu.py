@@ -242,3 +242,13 @@ def test_gpu_mttkrp_rank128(t3):
         out = sp.mttkrp(cs, mats_g, mode)
         ref = sp.mttkrp_stream(t3, mats_c, mode)
         assert (out.cpu() - ref).abs().max().item() < 1e-8
+
+
+def test_gpu_event_timer():
+    from splatt_amd.utils.timers import CudaEventTimer
+    t = CudaEventTimer()
+    x = torch.rand(1 << 20, device="cuda")
+    with t.time():
+        for _ in range(4):
+            x = x * 1.0001
+    assert t.elapsed_ms() > 0.0
