@@ -103,7 +103,9 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             fit = old_fit = float(ck["fit"])
             trace = list(ck.get("fit_trace", []))
 
+    import time as _time
     for it in range(it0, opts.max_iters):
+        _t0 = _time.perf_counter()
         for m in range(nm):
             mb = buf[: dims[m]]
             mttkrp(cs, factors, m, out=mb, nthreads=opts.nthreads)
@@ -133,7 +135,8 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
         trace.append(fit)
         niters = it + 1
         if opts.verbose:
-            print(f"  its = {it + 1} fit = {fit:.5f} delta = {fit - old_fit:+.4e}")
+            print(f"  its = {it + 1} ({_time.perf_counter() - _t0:.3f}s) "
+                  f"fit = {fit:.5f} delta = {fit - old_fit:+.4e}")
         if opts.checkpoint_path and (it + 1) % opts.checkpoint_every == 0:
             import os
             tmp = opts.checkpoint_path + ".tmp"

@@ -346,9 +346,14 @@ def grid_cpd_als(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
     opts = opts or CpdOptions()
     st = grid_cpd_init(shard_cs, dec, rank_f, opts)
     trace = []
+    import time as _time
     for it in range(opts.max_iters):
+        _t0 = _time.perf_counter()
         fit = grid_cpd_step(st, it)
         trace.append(fit)
+        if opts.verbose and (_world() == 1 or dist.get_rank() == 0):
+            print(f"  its = {it + 1} ({_time.perf_counter() - _t0:.3f}s) "
+                  f"fit = {fit:.5f}", flush=True)
         if it > 0 and abs(fit - st.old_fit) < opts.tolerance:
             break
     # post-process with GLOBAL column norms (chunked factors)
