@@ -186,7 +186,31 @@ def _build_csf_device(t: SpTensor, perm: List[int],
         keys = t.inds[perm[level]].index_select(0, order)
         order = order.index_select(0, torch.argsort(keys, stable=True))
     stage_meta = None
-    if flat_only and (gather_tiles > 1 or stage_rank > 0):
+    cache_mb = 0
+    if flat_only:
+        import os as _os2
+        cache_mb = int(_os2.environ.get("SPLATT_CACHE_TILE_MB", "0"))
+    if flat_only and cache_mb > 0 and stage_rank > 0:
+        # EXPERIMENTAL 3-D cache tiling for HBM-bound shapes: bucket by a
+        # cell over ALL modes with per-mode factor windows <= cache_mb, so
+        # every stream (output atomics + both gathers) works in an
+        # L3-resident window. Output runs shrink to nnz/(ncells*root_dim);
+        # the kernel is unchanged (atomic per run).
+        vbytes = t.vals.element_size()
+        win = cache_mb * 1024 * 1024
+        cell = torch.zeros(nnz, dtype=torch.int64, device=dev)
+        ncells = 1
+        for l in range(nm):
+            d = t.dims[perm[l]]
+            tl = max(1, (d * stage_rank * vbytes + win - 1) // win)
+            ch = (d + tl - 1) // tl
+            b = torch.div(t.inds[perm[l]].index_select(0, order), ch,
+                          rounding_mode="floor")
+            cell = cell * tl + b
+            ncells *= tl
+        order = order.index_select(0, torch.argsort(cell, stable=True))
+        stage_rank = 0  # no LDS staging on this sort order
+    elif flat_only and (gather_tiles > 1 or stage_rank > 0):
         big = max(range(1, nm), key=lambda l: t.dims[perm[l]])
         if stage_rank > 0:
             # LDS-staging buckets: rows-per-bucket sized to the LDS budget.
