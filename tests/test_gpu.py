@@ -252,3 +252,12 @@ def test_gpu_event_timer():
         for _ in range(4):
             x = x * 1.0001
     assert t.elapsed_ms() > 0.0
+
+
+@pytest.mark.parametrize("rank", [16, 32, 64])
+def test_gpu_gram_f32_mfma(rank):
+    from splatt_amd.ops.dense import gram
+    A = (torch.rand(20011, rank, dtype=torch.float32).cuda() - 0.5)
+    G = gram(A)
+    ref = A.double().T @ A.double()
+    assert (G.double() - ref).abs().max().item() < 0.05  # f32 accumulation
