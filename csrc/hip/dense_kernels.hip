@@ -71,11 +71,19 @@ template <> struct MfmaAcc<double> {
   static __device__ __forceinline__ type mfma(double a, double b, type c) {
     return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
   }
+  // f64 16x16x4 C/D row map (probed on gfx950): row = lane/16 + 4*reg
+  static __device__ __forceinline__ int crow(int lane, int i) {
+    return (lane >> 4) + 4 * i;
+  }
 };
 template <> struct MfmaAcc<float> {
   using type = __attribute__((ext_vector_type(4))) float;
   static __device__ __forceinline__ type mfma(float a, float b, type c) {
     return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+  // f32 16x16x4 uses the standard map: row = (lane/16)*4 + reg
+  static __device__ __forceinline__ int crow(int lane, int i) {
+    return (lane >> 4) * 4 + i;
   }
 };
 
@@ -123,7 +131,7 @@ gram_mfma_kern(const V * __restrict__ A, int64_t n, int64_t rows_per_blk,
     for (int bj = bi; bj < NB; ++bj, ++p) {
       #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int row = bi * 16 + (lane >> 4) + 4 * i;
+        const int row = bi * 16 + MfmaAcc<V>::crow(lane, i);
         const int col = bj * 16 + c;
         atomic_add_g(&G[row * F + col], acc[p][i]);
         if (bi != bj) atomic_add_g(&G[col * F + row], acc[p][i]);
