@@ -73,6 +73,7 @@ class GridDecomp:
     chunkn: List[int]
     global_dims: List[int]
     layer_groups: Dict[int, Optional[object]] = field(default_factory=dict)
+    layer_ranks: Dict[int, list] = field(default_factory=dict)
 
     @staticmethod
     def create_fine(global_dims: List[int],
@@ -92,6 +93,7 @@ class GridDecomp:
                        nnz_part=nnz_part)
         for m in range(len(global_dims)):
             d.layer_groups[m] = None if world == 1 else dist.group.WORLD
+            d.layer_ranks[m] = list(range(world))
         return d
 
     @staticmethod
@@ -137,6 +139,7 @@ class GridDecomp:
         for m in range(nm):
             if self.grid[m] == 1:
                 self.layer_groups[m] = dist.group.WORLD
+                self.layer_ranks[m] = list(range(world))
                 continue
             mine = None
             for v in range(self.grid[m]):
@@ -145,6 +148,7 @@ class GridDecomp:
                 g = dist.new_group(ranks) if len(ranks) > 1 else "solo"
                 if v == self.coords[m]:
                     mine = g
+                    self.layer_ranks[m] = ranks
             self.layer_groups[m] = mine
 
     def repl(self, m: int) -> int:
@@ -188,6 +192,55 @@ class FineDecomp(GridDecomp):
 def _ar(t: torch.Tensor, group=None, op=None) -> None:
     if _world() > 1 and group != "solo":
         dist.all_reduce(t, op=op or dist.ReduceOp.SUM, group=group)
+
+
+def _backend_is_nccl() -> bool:
+    try:
+        return dist.get_backend() == "nccl"
+    except Exception:  # noqa: BLE001
+        return False
+
+
+def _reduce_scatter_rows(full: torch.Tensor, lo: int, hi: int,
+                         group, gsize: int) -> torch.Tensor:
+    """Sum `full` across the group and return this rank's owned row block
+    [lo, hi). RCCL: true ncclReduceScatter over equal padded chunks
+    (SURVEY.md §2.4 mapping); gloo (CPU tests): all-reduce + slice, which
+    is the same math in a different summation order."""
+    n, F = full.shape
+    if not _backend_is_nccl():
+        dist.all_reduce(full, group=group)
+        return full[lo:hi]
+    per = (n + gsize - 1) // gsize
+    if n == per * gsize and (hi - lo) == per:
+        out = torch.empty(per, F, dtype=full.dtype, device=full.device)
+        dist.reduce_scatter_tensor(out, full.contiguous(), group=group)
+        return out[: hi - lo]
+    padded = torch.zeros(per * gsize, F, dtype=full.dtype, device=full.device)
+    padded[:n] = full
+    out = torch.empty(per, F, dtype=full.dtype, device=full.device)
+    dist.reduce_scatter_tensor(out, padded, group=group)
+    return out[: hi - lo]
+
+
+def _all_gather_rows(own: torch.Tensor, n: int, group, gsize: int,
+                     out: torch.Tensor) -> None:
+    """Concatenate the group's owned row blocks back into `out` (n x F)."""
+    per = (n + gsize - 1) // gsize
+    F = own.shape[1]
+    src = own
+    if own.shape[0] != per:
+        src = torch.zeros(per, F, dtype=own.dtype, device=own.device)
+        src[: own.shape[0]] = own
+    if _backend_is_nccl():
+        buf = torch.empty(per * gsize, F, dtype=own.dtype, device=own.device)
+        dist.all_gather_into_tensor(buf, src.contiguous(), group=group)
+    else:
+        parts = [torch.empty(per, F, dtype=own.dtype, device=own.device)
+                 for _ in range(gsize)]
+        dist.all_gather(parts, src.contiguous(), group=group)
+        buf = torch.cat(parts, 0)
+    out.copy_(buf[:n])
 
 
 @dataclass
@@ -249,22 +302,64 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
     dtype = st.buf.dtype
     F = st.factors[0].shape[1]
 
+    import os as _os
+    use_rsag = _os.environ.get("SPLATT_NO_RSAG") != "1"
     for m in range(nm):
         mb = st.buf[: dec.chunkn[m]]
         with tm("MTTKRP"):
             mttkrp(st.cs, st.factors, m, out=mb)
-        work = None
         group = dec.layer_groups.get(m)
-        if _world() > 1 and group != "solo" and dec.repl(m) > 1:
-            with tm("COMM-POST"):
-                work = dist.all_reduce(mb, group=group, async_op=True)
-        # Gram product + inverse are independent of mb: compute under comm
+        distributed = _world() > 1 and group != "solo" and dec.repl(m) > 1
+        # Gram product + inverse are independent of mb
         with tm("SOLVE"):
             G = torch.ones(F, F, dtype=dtype, device=dev)
             for o in range(nm):
                 if o != m:
                     G *= st.grams[o]
             Ginv = spd_inverse(G)
+        if distributed and use_rsag:
+            # the SURVEY §2.4 mapping: reduce-scatter partial rows to
+            # contiguous owners, solve/normalize/gram ONLY owned rows,
+            # all-gather the updated blocks (replaces alltoallv pair)
+            gsize = dec.repl(m)
+            my = dec.layer_ranks[m].index(dist.get_rank())
+            nrows = dec.chunkn[m]
+            per = (nrows + gsize - 1) // gsize
+            lo = min(my * per, nrows)
+            hi = min(lo + per, nrows)
+            with tm("COMM-RS"):
+                own_mb = _reduce_scatter_rows(mb, lo, hi, group, gsize)
+            A_own = own_mb @ Ginv
+            # lambda over GLOBAL rows (owned rows are globally unique)
+            if it == 0:
+                s = A_own.square().sum(dim=0)
+                _ar(s)
+                lam = s.sqrt()
+            else:
+                lam = A_own.abs().amax(dim=0) if A_own.numel() else                     torch.zeros(F, dtype=dtype, device=dev)
+                _ar(lam, op=dist.ReduceOp.MAX)
+                lam = lam.clamp_(min=1.0)
+            lam = torch.where(lam == 0, torch.ones_like(lam), lam)
+            A_own = A_own / lam
+            g = gram(A_own) if A_own.numel() else                 torch.zeros(F, F, dtype=dtype, device=dev)
+            _ar(g)
+            A = torch.empty(nrows, F, dtype=dtype, device=dev)
+            with tm("COMM-AG"):
+                _all_gather_rows(A_own, nrows, group, gsize, A)
+            st.lam = lam
+            st.factors[m] = A
+            st.grams[m] = g
+            if m == nm - 1:
+                # fit inner from OWNED rows (each global row owned exactly
+                # once across the job): partial now, summed at fit time
+                part = (own_mb.double() * A_own.double()).sum(dim=0)                     if A_own.numel() else torch.zeros(F, dtype=torch.float64,
+                                                      device=dev)
+                st._rs_inner = part  # type: ignore[attr-defined]
+            continue
+        work = None
+        if distributed:
+            with tm("COMM-POST"):
+                work = dist.all_reduce(mb, group=group, async_op=True)
         if work is not None:
             with tm("COMM-WAIT"):
                 work.wait()
@@ -293,11 +388,17 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
         st.grams[m] = g
 
     mlast = nm - 1
-    inner_t = (st.buf[: dec.chunkn[mlast]].double()
-               * st.factors[mlast].double()).sum(dim=0) @ st.lam.double()
-    if dec.grid[mlast] > 1:
-        inner_t /= dec.repl(mlast)
-        _ar(inner_t)
+    rs_part = getattr(st, "_rs_inner", None)
+    if rs_part is not None:
+        _ar(rs_part)
+        inner_t = rs_part @ st.lam.double()
+        st._rs_inner = None  # type: ignore[attr-defined]
+    else:
+        inner_t = (st.buf[: dec.chunkn[mlast]].double()
+                   * st.factors[mlast].double()).sum(dim=0) @ st.lam.double()
+        if dec.grid[mlast] > 1:
+            inner_t /= dec.repl(mlast)
+            _ar(inner_t)
     inner = float(inner_t)
     Gall = torch.ones(F, F, dtype=dtype, device=dev)
     for o in range(nm):
