@@ -261,3 +261,23 @@ def test_gpu_gram_f32_mfma(rank):
     G = gram(A)
     ref = A.double().T @ A.double()
     assert (G.double() - ref).abs().max().item() < 0.05  # f32 accumulation
+
+
+def test_gpu_rccl_rs_ag_helpers(tmp_path):
+    """Exercise the RCCL collective signatures used by the owned-row
+    update path on a 1-rank NCCL group (validates the nccl code path the
+    multi-GPU round-end run takes)."""
+    import torch.distributed as dist
+    from splatt_amd.parallel.grid import _all_gather_rows, _reduce_scatter_rows
+    dist.init_process_group(
+        "nccl", init_method=f"file://{tmp_path}/store1", rank=0, world_size=1)
+    try:
+        full = torch.rand(103, 16, dtype=torch.float64).cuda()
+        want = full.clone()
+        own = _reduce_scatter_rows(full, 0, 103, dist.group.WORLD, 1)
+        assert torch.allclose(own, want)
+        out = torch.empty(103, 16, dtype=torch.float64).cuda()
+        _all_gather_rows(own, 103, dist.group.WORLD, 1, out)
+        assert torch.allclose(out, want)
+    finally:
+        dist.destroy_process_group()
