@@ -202,25 +202,25 @@ def _backend_is_nccl() -> bool:
 
 
 def _reduce_scatter_rows(full: torch.Tensor, lo: int, hi: int,
-                         group, gsize: int) -> torch.Tensor:
-    """Sum `full` across the group and return this rank's owned row block
-    [lo, hi). RCCL: true ncclReduceScatter over equal padded chunks
+                         group, gsize: int):
+    """Sum `full` across the group and return (owned_rows, work) for this
+    rank's block [lo, hi); `work` is an async handle (wait before reading
+    owned_rows). RCCL: true ncclReduceScatter over equal padded chunks
     (SURVEY.md §2.4 mapping); gloo (CPU tests): all-reduce + slice, which
     is the same math in a different summation order."""
     n, F = full.shape
     if not _backend_is_nccl():
-        dist.all_reduce(full, group=group)
-        return full[lo:hi]
+        work = dist.all_reduce(full, group=group, async_op=True)
+        return full[lo:hi], work
     per = (n + gsize - 1) // gsize
-    if n == per * gsize and (hi - lo) == per:
-        out = torch.empty(per, F, dtype=full.dtype, device=full.device)
-        dist.reduce_scatter_tensor(out, full.contiguous(), group=group)
-        return out[: hi - lo]
-    padded = torch.zeros(per * gsize, F, dtype=full.dtype, device=full.device)
-    padded[:n] = full
+    src = full.contiguous()
+    if n != per * gsize:
+        src = torch.zeros(per * gsize, F, dtype=full.dtype,
+                          device=full.device)
+        src[:n] = full
     out = torch.empty(per, F, dtype=full.dtype, device=full.device)
-    dist.reduce_scatter_tensor(out, padded, group=group)
-    return out[: hi - lo]
+    work = dist.reduce_scatter_tensor(out, src, group=group, async_op=True)
+    return out[: hi - lo], work
 
 
 def _all_gather_rows(own: torch.Tensor, n: int, group, gsize: int,
@@ -310,17 +310,11 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             mttkrp(st.cs, st.factors, m, out=mb)
         group = dec.layer_groups.get(m)
         distributed = _world() > 1 and group != "solo" and dec.repl(m) > 1
-        # Gram product + inverse are independent of mb
-        with tm("SOLVE"):
-            G = torch.ones(F, F, dtype=dtype, device=dev)
-            for o in range(nm):
-                if o != m:
-                    G *= st.grams[o]
-            Ginv = spd_inverse(G)
         if distributed and use_rsag:
             # the SURVEY §2.4 mapping: reduce-scatter partial rows to
             # contiguous owners, solve/normalize/gram ONLY owned rows,
-            # all-gather the updated blocks (replaces alltoallv pair)
+            # all-gather the updated blocks (replaces alltoallv pair).
+            # The Gram Hadamard + inverse run UNDER the async collective.
             gsize = dec.repl(m)
             my = dec.layer_ranks[m].index(dist.get_rank())
             nrows = dec.chunkn[m]
@@ -328,7 +322,16 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             lo = min(my * per, nrows)
             hi = min(lo + per, nrows)
             with tm("COMM-RS"):
-                own_mb = _reduce_scatter_rows(mb, lo, hi, group, gsize)
+                own_mb, work = _reduce_scatter_rows(mb, lo, hi, group, gsize)
+            with tm("SOLVE"):
+                G = torch.ones(F, F, dtype=dtype, device=dev)
+                for o in range(nm):
+                    if o != m:
+                        G *= st.grams[o]
+                Ginv = spd_inverse(G)
+            if work is not None:
+                with tm("COMM-WAIT"):
+                    work.wait()
             A_own = own_mb @ Ginv
             # lambda over GLOBAL rows (owned rows are globally unique)
             if it == 0:
@@ -360,6 +363,12 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
         if distributed:
             with tm("COMM-POST"):
                 work = dist.all_reduce(mb, group=group, async_op=True)
+        with tm("SOLVE"):
+            G = torch.ones(F, F, dtype=dtype, device=dev)
+            for o in range(nm):
+                if o != m:
+                    G *= st.grams[o]
+            Ginv = spd_inverse(G)
         if work is not None:
             with tm("COMM-WAIT"):
                 work.wait()
