@@ -274,7 +274,9 @@ def test_gpu_rccl_rs_ag_helpers(tmp_path):
     try:
         full = torch.rand(103, 16, dtype=torch.float64).cuda()
         want = full.clone()
-        own = _reduce_scatter_rows(full, 0, 103, dist.group.WORLD, 1)
+        own, work = _reduce_scatter_rows(full, 0, 103, dist.group.WORLD, 1)
+        if work is not None:
+            work.wait()
         assert torch.allclose(own, want)
         out = torch.empty(103, 16, dtype=torch.float64).cuda()
         _all_gather_rows(own, 103, dist.group.WORLD, 1, out)
