@@ -229,6 +229,8 @@ def main():
                 "nnz": nnz_global,
                 "cp_rank": rank_f,
                 "csf": args.csf,
+                "tiling": ("lds-staged buckets" if stage_rank else
+                           (f"gather-range x{gt}" if gt > 1 else "none")),
                 "parallelism": f"{args.decomp}-grid {dec.grid} x{world} (RCCL/xGMI)",
                 "fit": round(st.fit, 6),
             },
