@@ -80,3 +80,15 @@ def test_kruskal_utilities(small3):
     dref = torch.zeros(*small3.dims, dtype=torch.float64)
     dref.index_put_(tuple(small3.inds), small3.vals.double(), accumulate=True)
     assert abs(kruskal_innerprod(k, small3) - float((dense * dref).sum())) < 1e-6
+
+
+def test_checkpoint_every_interval(tmp_path, small3):
+    import torch
+    ck = str(tmp_path / "int.ckpt")
+    sp.cpd_als(small3, 6, sp.CpdOptions(max_iters=5, tolerance=0.0,
+                                        checkpoint_path=ck,
+                                        checkpoint_every=2))
+    state = torch.load(ck, weights_only=True)
+    # last multiple-of-2 iteration is it=3 (0-indexed), i.e. 4 iterations
+    assert state["iteration"] == 3
+    assert len(state["factors"]) == 3
