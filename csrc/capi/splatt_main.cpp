@@ -160,11 +160,16 @@ int main(int argc, char ** argv) {
   const std::string cmd = argv[1];
   int sub_argc = argc - 2;
   char ** sub_argv = argv + 2;
-  if (cmd == "cpd") return cmd_cpd(sub_argc, sub_argv);
-  if (cmd == "check") return cmd_check(sub_argc, sub_argv);
-  if (cmd == "convert") return cmd_convert(sub_argc, sub_argv);
-  if (cmd == "stats") return cmd_stats(sub_argc, sub_argv);
-  if (cmd == "bench") return cmd_bench(sub_argc, sub_argv);
+  try {
+    if (cmd == "cpd") return cmd_cpd(sub_argc, sub_argv);
+    if (cmd == "check") return cmd_check(sub_argc, sub_argv);
+    if (cmd == "convert") return cmd_convert(sub_argc, sub_argv);
+    if (cmd == "stats") return cmd_stats(sub_argc, sub_argv);
+    if (cmd == "bench") return cmd_bench(sub_argc, sub_argv);
+  } catch (const std::exception & e) {
+    std::fprintf(stderr, "splatt: %s\n", e.what());
+    return 1;
+  }
   usage();
   return 1;
 }

@@ -129,3 +129,10 @@ def test_csf_convert_in_memory(lib, small3):
     assert lib.splatt_csf_nnz(csf) == nnz
     lib.splatt_free_csf(csf, o)
     lib.splatt_free_opts(o)
+
+
+def test_cli_binary_missing_file():
+    r = subprocess.run([EXE, "stats", "/nonexistent.tns"],
+                       capture_output=True, text=True)
+    assert r.returncode == 1
+    assert "cannot open" in r.stderr
