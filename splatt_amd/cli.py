@@ -234,7 +234,11 @@ def main(argv=None) -> int:
     p.set_defaults(fn=cmd_reorder)
 
     args = ap.parse_args(argv)
-    return args.fn(args)
+    try:
+        return args.fn(args)
+    except (RuntimeError, OSError, ValueError) as e:
+        print(f"splatt: {e}", file=sys.stderr)
+        return 1
 
 
 if __name__ == "__main__":
