@@ -195,6 +195,13 @@ def _ar(t: torch.Tensor, group=None, op=None) -> None:
 
 
 def _backend_is_nccl() -> bool:
+    """True -> use the real reduce_scatter_tensor / all_gather_into_tensor
+    primitives (the RCCL path). SPLATT_FORCE_RS_PRIMS=1 forces this branch
+    under gloo (torch>=2.10 implements both on gloo) so CPU tests cover the
+    exact padding/ownership code the 8-GPU RCCL run executes."""
+    import os
+    if os.environ.get("SPLATT_FORCE_RS_PRIMS") == "1":
+        return True
     try:
         return dist.get_backend() == "nccl"
     except Exception:  # noqa: BLE001

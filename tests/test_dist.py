@@ -256,3 +256,74 @@ def test_world8_matches_single(tmp_path, grid):
         p.join(timeout=300)
         assert p.exitcode == 0
     assert abs(fit8 - k1.fit) < 1e-8, (fit8, k1.fit, grid)
+
+
+def _rsag_worker(rank, world, file_store, result_q):
+    # SPLATT_FORCE_RS_PRIMS: take the RCCL branch of _reduce_scatter_rows /
+    # _all_gather_rows (true reduce_scatter_tensor + all_gather_into_tensor,
+    # equal padded chunks) over gloo — the exact code the 8-GPU run executes.
+    os.environ["SPLATT_FORCE_RS_PRIMS"] = "1"
+    torch.distributed.init_process_group(
+        "gloo", init_method=f"file://{file_store}", rank=rank,
+        world_size=world)
+    try:
+        from splatt_amd.parallel.grid import (
+            GridDecomp, grid_cpd_als, _reduce_scatter_rows, _all_gather_rows)
+        from splatt_amd.parallel.dist_cpd import build_shard_csf
+
+        # primitive-level check, uneven rows (n % gsize != 0 -> padding)
+        for n in (10, 9, 7):
+            base = torch.arange(n * 4, dtype=torch.float64).reshape(n, 4)
+            full = base * (rank + 1)
+            ref = base * sum(r + 1 for r in range(world))  # all-reduce result
+            per = (n + world - 1) // world
+            lo = min(rank * per, n)
+            hi = min(lo + per, n)
+            own, work = _reduce_scatter_rows(full.clone(), lo, hi,
+                                             torch.distributed.group.WORLD,
+                                             world)
+            if work is not None:
+                work.wait()
+            assert torch.allclose(own, ref[lo:hi]), (n, rank)
+            out = torch.empty(n, 4, dtype=torch.float64)
+            _all_gather_rows(own, n, torch.distributed.group.WORLD, world, out)
+            assert torch.allclose(out, ref), (n, rank)
+
+        # end-to-end: grid CPD fit must equal the single-process fit.
+        # grid [2,1,2] at world 4: mode 1 has chunkn=30, layer size 4 ->
+        # per=8 with a 6-row last owner + 2 pad rows (uneven path).
+        t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+        dec = GridDecomp.create(list(DIMS), grid=[2, 1, 2])
+        shard = dec.localize(t)
+        cs = build_shard_csf(shard, list(DIMS), "two")
+        opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+        k = grid_cpd_als(cs, dec, RANK_F, opts)
+        if rank == 0:
+            result_q.put(("fit", k.fit))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_grid_cpd_true_rs_primitives(tmp_path):
+    """Force the RCCL reduce-scatter/all-gather tensor primitives (padded
+    equal chunks) under gloo and require the world-4 fit to equal the
+    single-process fit — covers the branch only nccl takes in production."""
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
+    opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
+    k1 = sp.cpd_als(t, RANK_F, opts)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    store = str(tmp_path / "store_rsag")
+    procs = [ctx.Process(target=_rsag_worker, args=(r, 4, store, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    tag, fit4 = q.get()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    assert tag == "fit"
+    assert abs(fit4 - k1.fit) < 1e-8, (fit4, k1.fit)
