@@ -217,7 +217,9 @@ def test_load_shard_and_write_factors(tmp_path):
     assert (tmp_path / "lambda.mat").exists()
 
 
-def _w8_worker(rank, world, file_store, result_q, grid):
+def _w8_worker(rank, world, file_store, result_q, grid, force_prims=False):
+    if force_prims:
+        os.environ["SPLATT_FORCE_RS_PRIMS"] = "1"
     torch.distributed.init_process_group(
         "gloo", init_method=f"file://{file_store}", rank=rank,
         world_size=world)
@@ -237,17 +239,23 @@ def _w8_worker(rank, world, file_store, result_q, grid):
 
 
 @pytest.mark.timeout(600)
-@pytest.mark.parametrize("grid", [[1, 1, 8], [2, 1, 4]])
-def test_world8_matches_single(tmp_path, grid):
-    """The round-end 8-rank topologies (coarse + medium), over gloo."""
+@pytest.mark.parametrize("grid,force_prims",
+                         [([1, 1, 8], False), ([2, 1, 4], False),
+                          ([1, 1, 8], True)])
+def test_world8_matches_single(tmp_path, grid, force_prims):
+    """The round-end 8-rank topologies (coarse + medium) over gloo; the
+    coarse one also with the true RCCL reduce-scatter/all-gather
+    primitives forced (SPLATT_FORCE_RS_PRIMS) — the exact SCALE_r
+    configuration of the driver."""
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
     k1 = sp.cpd_als(t, RANK_F, sp.CpdOptions(max_iters=3, tolerance=0.0,
                                              seed=SEED))
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    store = str(tmp_path / f"store8_{grid[0]}_{grid[2]}")
-    procs = [ctx.Process(target=_w8_worker, args=(r, 8, store, q, grid))
+    store = str(tmp_path / f"store8_{grid[0]}_{grid[2]}_{force_prims}")
+    procs = [ctx.Process(target=_w8_worker,
+                         args=(r, 8, store, q, grid, force_prims))
              for r in range(8)]
     for p in procs:
         p.start()
@@ -255,7 +263,7 @@ def test_world8_matches_single(tmp_path, grid):
     for p in procs:
         p.join(timeout=300)
         assert p.exitcode == 0
-    assert abs(fit8 - k1.fit) < 1e-8, (fit8, k1.fit, grid)
+    assert abs(fit8 - k1.fit) < 1e-8, (fit8, k1.fit, grid, force_prims)
 
 
 def _rsag_worker(rank, world, file_store, result_q):
