@@ -391,6 +391,14 @@ void splatt_hip_mttkrp_flat_f64(
 void splatt_hip_mttkrp_flat_f32(
     const int32_t*, const int32_t* const*, const float* const*,
     const float*, int64_t, float*, int, int, void*);
+// deterministic flat kernel, csrc/hip/mttkrp_det.hip
+int64_t splatt_hip_flat_det_ws(int64_t, int);
+int splatt_hip_mttkrp_flat_det_f64(
+    const int32_t*, const int32_t* const*, const double* const*,
+    const double*, int64_t, double*, double*, int64_t, int, int, void*);
+int splatt_hip_mttkrp_flat_det_f32(
+    const int32_t*, const int32_t* const*, const float* const*,
+    const float*, int64_t, float*, float*, int64_t, int, int, void*);
 // dense kernels, csrc/hip/dense_kernels.hip
 void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
 void splatt_hip_gram_f32(const float*, int64_t, int, float*, void*);
@@ -489,6 +497,45 @@ static void py_gpu_mttkrp_flat(Tensor key, std::vector<Tensor> idx,
   }
 }
 
+// deterministic flat MTTKRP (csrc/hip/mttkrp_det.hip): plain stores for
+// walker-interior key runs + ordered fixup of boundary partials in `side`
+static void py_gpu_mttkrp_flat_det(Tensor key, std::vector<Tensor> idx,
+                                   std::vector<Tensor> mats, Tensor vals,
+                                   Tensor out, Tensor side, int64_t stream) {
+  const int nother = (int)idx.size();
+  TORCH_CHECK(nother >= 2 && nother <= 4,
+              "deterministic kernel supports 3..5 modes");
+  TORCH_CHECK((int)mats.size() == nother);
+  const int rank = (int)mats[0].size(1);
+  const int64_t nnz = vals.numel();
+  const int32_t * ip[8] = {};
+  for (int t = 0; t < nother; ++t) ip[t] = idx[t].data_ptr<int32_t>();
+  int rc;
+  if (vals.scalar_type() == torch::kFloat64) {
+    const double * mp[8] = {};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
+    rc = splatt_hip_mttkrp_flat_det_f64(key.data_ptr<int32_t>(), ip, mp,
+                                        vals.data_ptr<double>(), nnz,
+                                        out.data_ptr<double>(),
+                                        side.data_ptr<double>(),
+                                        side.numel(), rank, nother,
+                                        (void*)stream);
+  } else {
+    const float * mp[8] = {};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+    rc = splatt_hip_mttkrp_flat_det_f32(key.data_ptr<int32_t>(), ip, mp,
+                                        vals.data_ptr<float>(), nnz,
+                                        out.data_ptr<float>(),
+                                        side.data_ptr<float>(),
+                                        side.numel(), rank, nother,
+                                        (void*)stream);
+  }
+  TORCH_CHECK(rc != -1, "deterministic kernel requires rank in "
+                        "{4,8,16,32,64} and <= 5 modes");
+  TORCH_CHECK(rc == 0, "deterministic workspace too small (", side.numel(),
+              " elems < ", splatt_hip_flat_det_ws(nnz, rank), ")");
+}
+
 // thin wrappers: Python passes contiguous CUDA tensors + stream handle
 template <typename V>
 static void gpu_mttkrp3_t(int which, Tensor fptr0, py::object fids0, Tensor fptr1,
@@ -548,6 +595,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gpu_mttkrp3", &py_gpu_mttkrp3, "3-mode CSF MTTKRP HIP kernels");
   m.def("gpu_mttkrp_flat", &py_gpu_mttkrp_flat,
         "flat expanded-CSF MTTKRP HIP kernel (3..5 modes)");
+  m.def("gpu_mttkrp_flat_det", &py_gpu_mttkrp_flat_det,
+        "bitwise-deterministic flat MTTKRP (depth-0 streams, spec ranks)");
+  m.def("flat_det_ws_elems", &splatt_hip_flat_det_ws,
+        "workspace elements required by gpu_mttkrp_flat_det");
   m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
   m.def("gpu_spd_inverse", [](Tensor G, Tensor Ginv, int64_t stream) {
     const int F = (int)G.size(0);

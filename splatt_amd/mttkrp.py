@@ -88,6 +88,41 @@ def _stage_blocks(c: Csf) -> dict:
     return blocks
 
 
+def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
+                    out: torch.Tensor) -> None:
+    """Bitwise-deterministic device MTTKRP (csrc/hip/mttkrp_det.hip):
+    plain stores for walker-interior output runs + a walker-ordered fixup
+    of span-boundary partials. Needs a depth-0 (root-sorted) stream —
+    every key contiguous — i.e. the default ALLMODE policy; and a spec
+    rank. The reference's reproducibility answer is its serial
+    mttkrp_stream (src/mttkrp.c:1697); this keeps the property on-GPU."""
+    nm = c.nmodes
+    rank = int(mats[0].shape[1])
+    if depth != 0:
+        raise ValueError(
+            "deterministic MTTKRP needs a root-sorted (depth-0) stream for "
+            f"this mode; got depth {depth} — build with csf_alloc='all'")
+    if rank not in (4, 8, 16, 32, 64) or nm > 5:
+        raise ValueError(
+            f"deterministic MTTKRP supports ranks 4/8/16/32/64 and <=5 "
+            f"modes, got rank {rank}, {nm} modes")
+    key = c.ancestor_expand(depth)
+    idx, ms = [], []
+    for l in range(nm):
+        if l == depth:
+            continue
+        idx.append(c.ancestor_expand(l))
+        ms.append(mats[c.dim_perm[l]].contiguous())
+    need = native().flat_det_ws_elems(c.nnz, rank)
+    ws = getattr(c, "_det_ws", None)
+    if (ws is None or ws.numel() < need or ws.dtype != c.vals.dtype
+            or ws.device != c.device):
+        ws = torch.empty(need, dtype=c.vals.dtype, device=c.device)
+        object.__setattr__(c, "_det_ws", ws)
+    stream = torch.cuda.current_stream().cuda_stream
+    native().gpu_mttkrp_flat_det(key, idx, ms, c.vals, out, ws, stream)
+
+
 def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
                      out: torch.Tensor) -> None:
     """Flat expanded-CSF kernel (see csrc/hip/mttkrp_flat.hip): per-nnz
@@ -128,12 +163,15 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
 
 def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
            out: Optional[torch.Tensor] = None,
-           nthreads: int = 0, alg: str = "flat") -> torch.Tensor:
+           nthreads: int = 0, alg: str = "flat",
+           deterministic: Optional[bool] = None) -> torch.Tensor:
     """MTTKRP for output `mode`; `mats` indexed by tensor mode.
 
     Device algorithms: 'flat' (default, expanded-CSF streaming kernel) or
     'csf' (hierarchical fiber-walk kernels, 3-mode only) — the reference
     keeps multiple MTTKRP algorithms selectable the same way (bench.c).
+    `deterministic` (or SPLATT_DETERMINISTIC=1) selects the
+    bitwise-reproducible device kernel (no atomics; ALLMODE + spec ranks).
     """
     if isinstance(src, CsfSet):
         c = src.csfs[src.mode_csf[mode]]
@@ -163,7 +201,11 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
         if native().hip_arch() != 950:
             raise RuntimeError("HIP kernels not built for gfx950")
         out.zero_()
-        if alg == "flat":
+        if deterministic is None:
+            deterministic = os.environ.get("SPLATT_DETERMINISTIC") == "1"
+        if deterministic:
+            _gpu_mttkrp_det(c, depth, mats, out)
+        elif alg == "flat":
             _gpu_mttkrp_flat(c, depth, mats, out)
         else:
             _gpu_mttkrp_csf(c, depth, mats, mode, out)

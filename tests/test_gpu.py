@@ -283,3 +283,72 @@ def test_gpu_rccl_rs_ag_helpers(tmp_path):
         assert torch.allclose(out, want)
     finally:
         dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("rank", [16, 32])
+def test_gpu_deterministic_matches_oracle(t3, rank):
+    """Deterministic (atomic-free) kernel vs the CPU oracle, all modes."""
+    mats_c = make_mats(t3.dims, rank)
+    mats_g = [m.cuda() for m in mats_c]
+    cs = sp.csf_alloc(t3.to("cuda"), "all")
+    for mode in range(3):
+        out = sp.mttkrp(cs, mats_g, mode, deterministic=True)
+        ref = sp.mttkrp_stream(t3, mats_c, mode)
+        err = (out.cpu() - ref).abs().max().item()
+        assert err < 1e-8, (rank, mode, err)
+
+
+def test_gpu_deterministic_bitwise_repeatable():
+    """Two runs (with unrelated GPU work in between) must be torch.equal —
+    the property the atomic path cannot give."""
+    t = sp.SpTensor.synthetic([40, 60, 3000], 400_000, seed=31)
+    mats = [m.cuda() for m in make_mats(t.dims, 16)]
+    cs = sp.csf_alloc(t.to("cuda"), "all")
+    for mode in range(3):
+        a = sp.mttkrp(cs, mats, mode, deterministic=True).clone()
+        _ = torch.rand(2048, 2048, device="cuda") @ \
+            torch.rand(2048, 2048, device="cuda")   # unrelated traffic
+        b = sp.mttkrp(cs, mats, mode, deterministic=True)
+        assert torch.equal(a, b), mode
+
+
+def test_gpu_deterministic_4mode():
+    t = sp.SpTensor.synthetic([90, 120, 80, 40], 150_000, seed=41)
+    mats_c = make_mats(t.dims, 32)
+    mats_g = [m.cuda() for m in mats_c]
+    cs = sp.csf_alloc(t.to("cuda"), "all")
+    for mode in range(4):
+        out = sp.mttkrp(cs, mats_g, mode, deterministic=True)
+        ref = sp.mttkrp_stream(t, mats_c, mode)
+        assert (out.cpu() - ref).abs().max().item() < 1e-8, mode
+        again = sp.mttkrp(cs, mats_g, mode, deterministic=True)
+        assert torch.equal(out, again)
+
+
+def test_gpu_deterministic_rejects_unsupported(t3):
+    mats = [m.cuda() for m in make_mats(t3.dims, 16)]
+    cs2 = sp.csf_alloc(t3.to("cuda"), "one")   # non-root output depths
+    deep_mode = next(m for m in range(3)
+                     if cs2.mode_depth[m] != 0)
+    with pytest.raises(ValueError, match="depth-0"):
+        sp.mttkrp(cs2, mats, deep_mode, deterministic=True)
+    mats7 = [m.cuda() for m in make_mats(t3.dims, 7)]
+    csa = sp.csf_alloc(t3.to("cuda"), "all")
+    with pytest.raises(ValueError, match="rank"):
+        sp.mttkrp(csa, mats7, 0, deterministic=True)
+
+
+def test_gpu_deterministic_cpd_repeatable():
+    """SPLATT_DETERMINISTIC=1 -> whole device CPD is run-to-run identical."""
+    import os
+    t = sp.SpTensor.synthetic([150, 200, 180], 90_000, seed=53)
+    os.environ["SPLATT_DETERMINISTIC"] = "1"
+    try:
+        opts = sp.CpdOptions(max_iters=4, tolerance=0.0, seed=9)
+        k1 = sp.cpd_als(sp.csf_alloc(t.to("cuda"), "all"), 16, opts)
+        k2 = sp.cpd_als(sp.csf_alloc(t.to("cuda"), "all"), 16, opts)
+    finally:
+        del os.environ["SPLATT_DETERMINISTIC"]
+    assert k1.fit == k2.fit
+    for a, b in zip(k1.factors, k2.factors):
+        assert torch.equal(a, b)
