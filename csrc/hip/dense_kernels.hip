@@ -251,7 +251,59 @@ spd_inverse_kern(const V * __restrict__ G, V * __restrict__ Ginv, int F) {
   }
 }
 
+// C[n,F] = A[n,F] @ B[F,F] with B staged in LDS — the CPD solve GEMM.
+// One thread per output element, a fixed F-length dot: no atomics, no
+// split-K, bitwise-deterministic (used when SPLATT_DETERMINISTIC=1; the
+// default path keeps the library GEMM). A[r,k] is broadcast across the F
+// column lanes of a row, so traffic is ~one read of A + one write of C.
+template <typename V, int F>
+__global__ void __launch_bounds__(256)
+rowsolve_kern(const V * __restrict__ A, const V * __restrict__ B,
+              V * __restrict__ C, int64_t n) {
+  __shared__ V Bs[F * F];
+  for (int i = threadIdx.x; i < F * F; i += blockDim.x) Bs[i] = B[i];
+  __syncthreads();
+  const int c = threadIdx.x % F;
+  const int64_t rstride = (int64_t)gridDim.x * (blockDim.x / F);
+  int64_t r = (int64_t)blockIdx.x * (blockDim.x / F) + threadIdx.x / F;
+  for (; r < n; r += rstride) {
+    V acc = (V)0;
+    #pragma unroll
+    for (int k = 0; k < F; ++k) acc += A[r * F + k] * Bs[k * F + c];
+    C[r * F + c] = acc;
+  }
+}
+
+template <typename V>
+int launch_rowsolve(const V * A, const V * B, V * C, int64_t n, int F,
+                    hipStream_t st) {
+  const int rows_pb = 256 / (F < 256 ? F : 256);
+  int64_t blocks = (n + rows_pb - 1) / rows_pb;
+  if (blocks > 8192) blocks = 8192;          // grid-stride beyond this
+  if (blocks < 1) blocks = 1;
+  dim3 g((uint32_t)blocks), b(256);
+  switch (F) {
+    case 4:  hipLaunchKernelGGL((rowsolve_kern<V, 4>),  g, b, 0, st, A, B, C, n); return 0;
+    case 8:  hipLaunchKernelGGL((rowsolve_kern<V, 8>),  g, b, 0, st, A, B, C, n); return 0;
+    case 16: hipLaunchKernelGGL((rowsolve_kern<V, 16>), g, b, 0, st, A, B, C, n); return 0;
+    case 32: hipLaunchKernelGGL((rowsolve_kern<V, 32>), g, b, 0, st, A, B, C, n); return 0;
+    case 64: hipLaunchKernelGGL((rowsolve_kern<V, 64>), g, b, 0, st, A, B, C, n); return 0;
+    default: return -1;
+  }
+}
+
 }  // namespace
+
+extern "C" int splatt_hip_rowsolve_f64(const double * A, const double * B,
+                                       double * C, int64_t n, int F,
+                                       void * stream) {
+  return launch_rowsolve<double>(A, B, C, n, F, (hipStream_t)stream);
+}
+extern "C" int splatt_hip_rowsolve_f32(const float * A, const float * B,
+                                       float * C, int64_t n, int F,
+                                       void * stream) {
+  return launch_rowsolve<float>(A, B, C, n, F, (hipStream_t)stream);
+}
 
 extern "C" void splatt_hip_gram_f64(const double * A, int64_t n, int F,
                                     double * G, void * stream) {

@@ -400,6 +400,10 @@ int splatt_hip_mttkrp_flat_det_f32(
     const int32_t*, const int32_t* const*, const float* const*,
     const float*, int64_t, float*, float*, int64_t, int, int, void*);
 // dense kernels, csrc/hip/dense_kernels.hip
+int splatt_hip_rowsolve_f64(const double*, const double*, double*, int64_t,
+                            int, void*);
+int splatt_hip_rowsolve_f32(const float*, const float*, float*, int64_t,
+                            int, void*);
 void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
 void splatt_hip_gram_f32(const float*, int64_t, int, float*, void*);
 void splatt_hip_spd_inverse_f64(const double*, double*, int, void*);
@@ -600,6 +604,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flat_det_ws_elems", &splatt_hip_flat_det_ws,
         "workspace elements required by gpu_mttkrp_flat_det");
   m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
+  m.def("gpu_rowsolve", [](Tensor A, Tensor B, Tensor C, int64_t stream) {
+    const int64_t n = A.size(0);
+    const int F = (int)A.size(1);
+    int rc;
+    if (A.scalar_type() == torch::kFloat64)
+      rc = splatt_hip_rowsolve_f64(A.data_ptr<double>(), B.data_ptr<double>(),
+                                   C.data_ptr<double>(), n, F, (void*)stream);
+    else
+      rc = splatt_hip_rowsolve_f32(A.data_ptr<float>(), B.data_ptr<float>(),
+                                   C.data_ptr<float>(), n, F, (void*)stream);
+    TORCH_CHECK(rc == 0, "rowsolve supports F in {4,8,16,32,64}, got ", F);
+  }, "C = A @ B (B is FxF, staged in LDS; bitwise-deterministic)");
   m.def("gpu_spd_inverse", [](Tensor G, Tensor Ginv, int64_t stream) {
     const int F = (int)G.size(0);
     if (G.scalar_type() == torch::kFloat64)

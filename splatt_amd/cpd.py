@@ -22,7 +22,7 @@ import torch
 from splatt_amd._ext import native
 from splatt_amd.csf import CsfSet, csf_alloc
 from splatt_amd.mttkrp import mttkrp
-from splatt_amd.ops.dense import gram, spd_inverse
+from splatt_amd.ops.dense import gram, solve_rows, spd_inverse
 from splatt_amd.sptensor import SpTensor
 
 
@@ -116,7 +116,7 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             # solve A * G = mttkrp  =>  G^T A^T = mttkrp^T (G symmetric)
             # F x F inverse once, then one well-shaped (n x F)(F x F) GEMM —
             # beats a trsm against an n-row RHS at these tiny F
-            A = mb @ spd_inverse(G)
+            A = solve_rows(mb, spd_inverse(G))
             lam = _normalize(A, it)
             factors[m] = A
             grams[m] = gram(A)
@@ -124,12 +124,14 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
         # fit from last mode's pre-solve MTTKRP output (reference trick):
         # <X,K> = sum_f lam_f * sum_i buf[i,f] * A_last[i,f]
         mlast = nm - 1
-        inner = float((buf[: dims[mlast]].double()
-                       * factors[mlast].double()).sum(dim=0) @ lam.double())
+        inner = float(((buf[: dims[mlast]].double()
+                        * factors[mlast].double()).sum(dim=0)
+                       * lam.double()).sum())
         Gall = ones.clone()
         for o in range(nm):
             Gall *= grams[o]
-        knorm = float(lam.double() @ Gall.double() @ lam.double())
+        knorm = float((Gall.double()
+                       * torch.outer(lam.double(), lam.double())).sum())
         residual = math.sqrt(max(0.0, norm_x + knorm - 2 * inner))
         fit = 1.0 - residual / math.sqrt(norm_x)
         trace.append(fit)

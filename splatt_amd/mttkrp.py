@@ -106,13 +106,26 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
         raise ValueError(
             f"deterministic MTTKRP supports ranks 4/8/16/32/64 and <=5 "
             f"modes, got rank {rank}, {nm} modes")
-    key = c.ancestor_expand(depth)
-    idx, ms = [], []
-    for l in range(nm):
-        if l == depth:
-            continue
-        idx.append(c.ancestor_expand(l))
-        ms.append(mats[c.dim_perm[l]].contiguous())
+    # the deterministic scheme needs every output key CONTIGUOUS in the
+    # stream. LDS-bucketed builds reorder the stream bucket-major, so a
+    # key-sorted copy is built once and cached (stable argsort -> the
+    # copy itself is deterministic).
+    streams = getattr(c, "_det_streams", None)
+    if streams is None:
+        streams = {}
+        object.__setattr__(c, "_det_streams", streams)
+    if depth not in streams:
+        key = c.ancestor_expand(depth)
+        idx = [c.ancestor_expand(l) for l in range(nm) if l != depth]
+        vals = c.vals
+        if not bool((key[1:] >= key[:-1]).all()):
+            perm = torch.argsort(key, stable=True)
+            key = key.index_select(0, perm).contiguous()
+            idx = [i.index_select(0, perm).contiguous() for i in idx]
+            vals = vals.index_select(0, perm).contiguous()
+        streams[depth] = (key, idx, vals)
+    key, idx, vals = streams[depth]
+    ms = [mats[c.dim_perm[l]].contiguous() for l in range(nm) if l != depth]
     need = native().flat_det_ws_elems(c.nnz, rank)
     ws = getattr(c, "_det_ws", None)
     if (ws is None or ws.numel() < need or ws.dtype != c.vals.dtype
@@ -120,7 +133,7 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
         ws = torch.empty(need, dtype=c.vals.dtype, device=c.device)
         object.__setattr__(c, "_det_ws", ws)
     stream = torch.cuda.current_stream().cuda_stream
-    native().gpu_mttkrp_flat_det(key, idx, ms, c.vals, out, ws, stream)
+    native().gpu_mttkrp_flat_det(key, idx, ms, vals, out, ws, stream)
 
 
 def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],

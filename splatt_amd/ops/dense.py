@@ -35,3 +35,19 @@ def spd_inverse(G: torch.Tensor) -> torch.Tensor:
         G + 1e-12 * G.diagonal().abs().max()
         * torch.eye(F, dtype=G.dtype, device=G.device))
     return torch.cholesky_inverse(L)
+
+
+def solve_rows(mb: torch.Tensor, Ginv: torch.Tensor) -> torch.Tensor:
+    """mb @ Ginv — the CPD solve GEMM. Default: library GEMM. With
+    SPLATT_DETERMINISTIC=1 on device (F in the spec set): the one-thread-
+    per-element HIP kernel (no split-K/atomics), so repeated runs are
+    bitwise identical."""
+    import os
+    F = mb.shape[1]
+    if (os.environ.get("SPLATT_DETERMINISTIC") == "1"
+            and mb.device.type == "cuda" and F in (4, 8, 16, 32, 64)):
+        C = torch.empty_like(mb)
+        native().gpu_rowsolve(mb.contiguous(), Ginv.contiguous(), C,
+                              torch.cuda.current_stream().cuda_stream)
+        return C
+    return mb @ Ginv

@@ -27,7 +27,7 @@ import torch.distributed as dist
 from splatt_amd.cpd import CpdOptions, Kruskal, seeded_init
 from splatt_amd.csf import CsfSet
 from splatt_amd.mttkrp import mttkrp
-from splatt_amd.ops.dense import gram, spd_inverse
+from splatt_amd.ops.dense import gram, solve_rows, spd_inverse
 from splatt_amd.sptensor import SpTensor
 
 
@@ -339,7 +339,7 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             if work is not None:
                 with tm("COMM-WAIT"):
                     work.wait()
-            A_own = own_mb @ Ginv
+            A_own = solve_rows(own_mb, Ginv)
             # lambda over GLOBAL rows (owned rows are globally unique)
             if it == 0:
                 s = A_own.square().sum(dim=0)
@@ -380,7 +380,7 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             with tm("COMM-WAIT"):
                 work.wait()
         with tm("SOLVE"):
-            A = mb @ Ginv
+            A = solve_rows(mb, Ginv)
         # lambda over GLOBAL rows of mode m
         if it == 0:
             s = A.square().sum(dim=0)
