@@ -274,6 +274,22 @@ rowsolve_kern(const V * __restrict__ A, const V * __restrict__ B,
   }
 }
 
+// Deterministic G = A^T A: one thread per (i,j) output pair, a serial
+// dot over all n rows — no cross-block partials, no atomics. ~n*F^2/256
+// serial FMAs per thread (~25 us at n=29k, F=16): the determinism-mode
+// companion of the MFMA gram (whose per-wave partials land atomically).
+template <typename V>
+__global__ void __launch_bounds__(256)
+gram_det_kern(const V * __restrict__ A, int64_t n, int F,
+              V * __restrict__ G) {
+  const int e = blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= F * F) return;
+  const int i = e / F, j = e % F;
+  V acc = (V)0;
+  for (int64_t r = 0; r < n; ++r) acc += A[r * F + i] * A[r * F + j];
+  G[e] = acc;
+}
+
 template <typename V>
 int launch_rowsolve(const V * A, const V * B, V * C, int64_t n, int F,
                     hipStream_t st) {
@@ -293,6 +309,19 @@ int launch_rowsolve(const V * A, const V * B, V * C, int64_t n, int F,
 }
 
 }  // namespace
+
+extern "C" void splatt_hip_gram_det_f64(const double * A, int64_t n, int F,
+                                        double * G, void * stream) {
+  const int blocks = (F * F + 255) / 256;
+  hipLaunchKernelGGL((gram_det_kern<double>), dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, A, n, F, G);
+}
+extern "C" void splatt_hip_gram_det_f32(const float * A, int64_t n, int F,
+                                        float * G, void * stream) {
+  const int blocks = (F * F + 255) / 256;
+  hipLaunchKernelGGL((gram_det_kern<float>), dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, A, n, F, G);
+}
 
 extern "C" int splatt_hip_rowsolve_f64(const double * A, const double * B,
                                        double * C, int64_t n, int F,

@@ -9,8 +9,16 @@ from splatt_amd._ext import native
 
 def gram(A: torch.Tensor) -> torch.Tensor:
     """A^T A for row-major (n x F). Uses the HIP gram kernel on device for
-    F <= 64 (rocBLAS picks a one-workgroup tile there); rocBLAS otherwise."""
+    F <= 64 (rocBLAS picks a one-workgroup tile there); rocBLAS otherwise.
+    SPLATT_DETERMINISTIC=1: the serial-dot kernel (the MFMA gram lands
+    per-wave partials with atomic adds, whose order varies run to run)."""
+    import os
     n, F = A.shape
+    if (os.environ.get("SPLATT_DETERMINISTIC") == "1"
+            and A.device.type == "cuda" and F <= 64 and A.is_contiguous()):
+        G = torch.empty(F, F, dtype=A.dtype, device=A.device)
+        native().gpu_gram_det(A, G, torch.cuda.current_stream().cuda_stream)
+        return G
     if A.device.type == "cuda" and F <= 64 and A.is_contiguous():
         G = torch.zeros(F, F, dtype=A.dtype, device=A.device)
         native().gpu_gram(A, G, torch.cuda.current_stream().cuda_stream)

@@ -387,3 +387,17 @@ def test_gpu_rowsolve_matches_matmul(rank):
         del os.environ["SPLATT_DETERMINISTIC"]
     assert (C - A @ B).abs().max().item() < 1e-10
     assert torch.equal(C, C2)
+
+
+def test_gpu_gram_det_matches():
+    import os
+    from splatt_amd.ops.dense import gram
+    A = torch.rand(29818, 16, dtype=torch.float64).cuda()
+    os.environ["SPLATT_DETERMINISTIC"] = "1"
+    try:
+        G = gram(A)
+        G2 = gram(A)
+    finally:
+        del os.environ["SPLATT_DETERMINISTIC"]
+    assert (G - A.T @ A).abs().max().item() < 1e-8
+    assert torch.equal(G, G2)
