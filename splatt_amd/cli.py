@@ -30,6 +30,11 @@ def _load(args) -> "sp.SpTensor":
 
 def cmd_cpd(args) -> int:
     import os
+    if getattr(args, "deterministic", False):
+        # bitwise-reproducible device kernels (docs/KERNELS.md); forces
+        # ALLMODE so every mode has a root-sorted stream
+        os.environ["SPLATT_DETERMINISTIC"] = "1"
+        args.csf = "all"
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world > 1:
         return _cmd_cpd_dist(args, world)
@@ -191,6 +196,9 @@ def main(argv=None) -> int:
     p.add_argument("--native", action="store_true",
                    help="use the C++ host driver (CPU reference path)")
     p.add_argument("--nowrite", action="store_true")
+    p.add_argument("--deterministic", action="store_true",
+                   help="bitwise-reproducible device CPD (atomic-free "
+                        "kernels, ~60%% throughput; implies --csf all)")
     p.add_argument("-v", "--verbose", action="count", default=0)
     p.set_defaults(fn=cmd_cpd)
 

@@ -209,3 +209,18 @@ def test_hgraph_fib_and_perm_matrix(small3):
     A = torch.rand(small3.dims[0], 4, dtype=torch.float64)
     B = ro.perm_matrix(A, perm.perms[0])
     assert torch.equal(B[perm.iperms[0]], A)
+
+
+def test_cli_cpd_deterministic_flag(tmp_path, capsys):
+    """--deterministic sets the env knob and forces ALLMODE; on CPU the
+    run itself is the (already deterministic) host path."""
+    import os
+    import splatt_amd as sp
+    from splatt_amd.cli import main
+    t = sp.SpTensor.synthetic([12, 10, 14], 300, seed=5)
+    f = tmp_path / "t.tns"
+    t.save(f)
+    rc = main(["cpd", str(f), "-r", "4", "-i", "2", "--device", "cpu",
+               "--deterministic", "--nowrite"])
+    assert rc == 0
+    assert os.environ.pop("SPLATT_DETERMINISTIC", None) == "1"
