@@ -274,19 +274,44 @@ rowsolve_kern(const V * __restrict__ A, const V * __restrict__ B,
   }
 }
 
-// Deterministic G = A^T A: one thread per (i,j) output pair, a serial
-// dot over all n rows — no cross-block partials, no atomics. ~n*F^2/256
-// serial FMAs per thread (~25 us at n=29k, F=16): the determinism-mode
-// companion of the MFMA gram (whose per-wave partials land atomically).
+// Deterministic G = A^T A, two stages, no atomics: stage 1 gives each
+// block a fixed row range and a PRIVATE F x F partial (plain stores);
+// stage 2 folds the partials in block order with one workgroup. The
+// reduction tree is a pure function of (n, F), so results are bitwise
+// run-to-run identical. (A one-workgroup serial version measured 5.3 ms
+// at n=29k — a latency-bound dependent chain; splitting across blocks
+// recovers the parallelism without reintroducing atomics.)
 template <typename V>
 __global__ void __launch_bounds__(256)
-gram_det_kern(const V * __restrict__ A, int64_t n, int F,
-              V * __restrict__ G) {
-  const int e = blockIdx.x * blockDim.x + threadIdx.x;
+gram_det_part_kern(const V * __restrict__ A, int64_t n, int F,
+                   int64_t rows_per_blk, V * __restrict__ Gpart) {
+  const int e = threadIdx.x;
   if (e >= F * F) return;
   const int i = e / F, j = e % F;
+  const int64_t r0 = (int64_t)blockIdx.x * rows_per_blk;
+  const int64_t r1 = r0 + rows_per_blk < n ? r0 + rows_per_blk : n;
   V acc = (V)0;
-  for (int64_t r = 0; r < n; ++r) acc += A[r * F + i] * A[r * F + j];
+  int64_t r = r0;
+  for (; r + 4 <= r1; r += 4) {
+    // four independent products -> the loads overlap the FMA chain
+    const V a0 = A[r * F + i] * A[r * F + j];
+    const V a1 = A[(r + 1) * F + i] * A[(r + 1) * F + j];
+    const V a2 = A[(r + 2) * F + i] * A[(r + 2) * F + j];
+    const V a3 = A[(r + 3) * F + i] * A[(r + 3) * F + j];
+    acc += (a0 + a1) + (a2 + a3);
+  }
+  for (; r < r1; ++r) acc += A[r * F + i] * A[r * F + j];
+  Gpart[(int64_t)blockIdx.x * F * F + e] = acc;
+}
+
+template <typename V>
+__global__ void __launch_bounds__(256)
+gram_det_fold_kern(const V * __restrict__ Gpart, int64_t nparts, int F,
+                   V * __restrict__ G) {
+  const int e = threadIdx.x;
+  if (e >= F * F) return;
+  V acc = (V)0;
+  for (int64_t b = 0; b < nparts; ++b) acc += Gpart[b * F * F + e];
   G[e] = acc;
 }
 
@@ -311,16 +336,22 @@ int launch_rowsolve(const V * A, const V * B, V * C, int64_t n, int F,
 }  // namespace
 
 extern "C" void splatt_hip_gram_det_f64(const double * A, int64_t n, int F,
+                                        double * Gpart, int64_t nparts,
                                         double * G, void * stream) {
-  const int blocks = (F * F + 255) / 256;
-  hipLaunchKernelGGL((gram_det_kern<double>), dim3(blocks), dim3(256), 0,
-                     (hipStream_t)stream, A, n, F, G);
+  const int64_t rpb = (n + nparts - 1) / nparts;
+  hipLaunchKernelGGL((gram_det_part_kern<double>), dim3((uint32_t)nparts),
+                     dim3(256), 0, (hipStream_t)stream, A, n, F, rpb, Gpart);
+  hipLaunchKernelGGL((gram_det_fold_kern<double>), dim3(1), dim3(256), 0,
+                     (hipStream_t)stream, Gpart, nparts, F, G);
 }
 extern "C" void splatt_hip_gram_det_f32(const float * A, int64_t n, int F,
+                                        float * Gpart, int64_t nparts,
                                         float * G, void * stream) {
-  const int blocks = (F * F + 255) / 256;
-  hipLaunchKernelGGL((gram_det_kern<float>), dim3(blocks), dim3(256), 0,
-                     (hipStream_t)stream, A, n, F, G);
+  const int64_t rpb = (n + nparts - 1) / nparts;
+  hipLaunchKernelGGL((gram_det_part_kern<float>), dim3((uint32_t)nparts),
+                     dim3(256), 0, (hipStream_t)stream, A, n, F, rpb, Gpart);
+  hipLaunchKernelGGL((gram_det_fold_kern<float>), dim3(1), dim3(256), 0,
+                     (hipStream_t)stream, Gpart, nparts, F, G);
 }
 
 extern "C" int splatt_hip_rowsolve_f64(const double * A, const double * B,

@@ -402,8 +402,10 @@ int splatt_hip_mttkrp_flat_det_f32(
 // dense kernels, csrc/hip/dense_kernels.hip
 int splatt_hip_rowsolve_f64(const double*, const double*, double*, int64_t,
                             int, void*);
-void splatt_hip_gram_det_f64(const double*, int64_t, int, double*, void*);
-void splatt_hip_gram_det_f32(const float*, int64_t, int, float*, void*);
+void splatt_hip_gram_det_f64(const double*, int64_t, int, double*, int64_t,
+                             double*, void*);
+void splatt_hip_gram_det_f32(const float*, int64_t, int, float*, int64_t,
+                             float*, void*);
 int splatt_hip_rowsolve_f32(const float*, const float*, float*, int64_t,
                             int, void*);
 void splatt_hip_gram_f64(const double*, int64_t, int, double*, void*);
@@ -606,16 +608,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flat_det_ws_elems", &splatt_hip_flat_det_ws,
         "workspace elements required by gpu_mttkrp_flat_det");
   m.def("gpu_gram", &py_gpu_gram, "G += A^T A (tall-skinny, F<=64)");
-  m.def("gpu_gram_det", [](Tensor A, Tensor G, int64_t stream) {
+  m.def("gpu_gram_det", [](Tensor A, Tensor Gpart, Tensor G,
+                           int64_t stream) {
     const int64_t n = A.size(0);
     const int F = (int)A.size(1);
+    const int64_t nparts = Gpart.numel() / (F * F);
+    TORCH_CHECK(nparts >= 1, "gram_det workspace too small");
     if (A.scalar_type() == torch::kFloat64)
       splatt_hip_gram_det_f64(A.data_ptr<double>(), n, F,
+                              Gpart.data_ptr<double>(), nparts,
                               G.data_ptr<double>(), (void*)stream);
     else
       splatt_hip_gram_det_f32(A.data_ptr<float>(), n, F,
+                              Gpart.data_ptr<float>(), nparts,
                               G.data_ptr<float>(), (void*)stream);
-  }, "G = A^T A, serial dot per output pair (bitwise-deterministic)");
+  }, "G = A^T A, per-block partials folded in fixed order "
+     "(bitwise-deterministic)");
   m.def("gpu_rowsolve", [](Tensor A, Tensor B, Tensor C, int64_t stream) {
     const int64_t n = A.size(0);
     const int F = (int)A.size(1);
