@@ -118,3 +118,20 @@ def test_det_walker_fixup_logic():
         # runs aligned to span boundaries
         key = np.repeat(np.arange(1200), 64)[:300_000]
         _check(key, rng.standard_normal(len(key)), rank, f"r{rank} aligned")
+
+
+def test_det_walker_fixup_random():
+    """Randomized shapes: random run-length distributions, ranks and
+    sizes — claim uniqueness + exact totals must hold for all."""
+    rng = np.random.default_rng(7)
+    for trial in range(30):
+        rank = int(rng.choice([4, 8, 16, 32, 64]))
+        n = int(rng.integers(1, 40_000))
+        style = trial % 3
+        if style == 0:
+            key = np.sort(rng.integers(0, max(1, n // 50) + 1, n))
+        elif style == 1:   # few giant runs
+            key = np.sort(rng.integers(0, 4, n))
+        else:              # zipf-ish run lengths
+            key = np.sort(rng.zipf(1.3, n) % max(2, n // 100))
+        _check(key, rng.standard_normal(n), rank, f"trial{trial} r{rank}")
