@@ -151,6 +151,12 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
         raise ValueError("mode dimensions above 2^32 are not supported "
                          "(CSF node ids are 32-bit; shard the mode first)")
     if t.device.type == "cuda":
+        # device streams store labels as SIGNED int32 (the HIP kernels'
+        # key/idx width); the host builder's unsigned ids reach 2^32
+        if any(d > 0x7FFFFFFF for d in t.dims):
+            raise ValueError(
+                "device builds support mode dimensions up to 2^31-1 "
+                "(int32 stream labels; shard the mode or build on CPU)")
         if lds_kb <= 0:
             import os
             lds_kb = int(os.environ.get("SPLATT_LDS_KB", "24"))

@@ -99,3 +99,21 @@ def test_frozen_csf_survives_device_move(small3):
     c2 = c.to("cpu")
     for l in range(3):
         assert torch.equal(c.ancestor_expand(l), c2.ancestor_expand(l))
+
+
+def test_device_build_dim_guard():
+    """Device streams are SIGNED int32: dims in [2^31, 2^32) must be
+    rejected by the device build (the unsigned host builder allows them
+    up to 2^32). The guard fires before any tensor work, so a stub with
+    just .dims/.device exercises it on CPU."""
+    from types import SimpleNamespace
+    import pytest
+    from splatt_amd.csf import build_csf
+    fake = SimpleNamespace(dims=[2, 2, 2 ** 31 + 5],
+                           device=SimpleNamespace(type="cuda"))
+    with pytest.raises(ValueError, match="2\\^31"):
+        build_csf(fake, [0, 1, 2])
+    fake_huge = SimpleNamespace(dims=[2, 2, 2 ** 32 + 5],
+                                device=SimpleNamespace(type="cpu"))
+    with pytest.raises(ValueError, match="2\\^32"):
+        build_csf(fake_huge, [0, 1, 2])
