@@ -2,8 +2,11 @@
 """GPU soak: random tensor/config trials, every MTTKRP checked against
 the CPU oracle; deterministic-mode trials additionally checked for
 torch.equal across two runs. Usage: gpu_soak.py [trials] [seed]."""
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
