@@ -27,14 +27,7 @@ def _mttkrp_giga(t, mats, mode):
     """GigaTensor-style: sparse unfolding X_(m) times the materialized
     Khatri-Rao product (reference mttkrp_giga, mttkrp.c:1604-1649)."""
     others = [m for m in range(t.nmodes) if m != mode]
-    ncols = 1
-    for m in others:
-        ncols *= t.dims[m]
-    col = torch.zeros(t.nnz, dtype=torch.int64, device=t.device)
-    for m in others:
-        col = col * t.dims[m] + t.inds[m]
-    X = torch.sparse_coo_tensor(torch.stack([t.inds[mode], col]), t.vals,
-                                (t.dims[mode], ncols)).coalesce()
+    X = t.unfold(mode)
     K = mats[others[0]]
     for m in others[1:]:
         K = (K.unsqueeze(1) * mats[m].unsqueeze(0)).reshape(-1, K.shape[1])

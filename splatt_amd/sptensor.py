@@ -51,6 +51,23 @@ class SpTensor:
     def normsq(self) -> float:
         return float(self.vals.double().square().sum())
 
+    def unfold(self, mode: int) -> torch.Tensor:
+        """Mode-`mode` matricization X_(m) as a sparse CSR matrix of shape
+        (dims[mode], prod(other dims)); column index combines the remaining
+        modes in increasing order (reference tt_unfold, sptensor.c:307-356,
+        which produces the CSR unfolding consumed by mttkrp_giga)."""
+        others = [m for m in range(self.nmodes) if m != mode]
+        ncols = 1
+        for m in others:
+            ncols *= self.dims[m]
+        col = torch.zeros(self.nnz, dtype=torch.int64, device=self.device)
+        for m in others:
+            col = col * self.dims[m] + self.inds[m]
+        coo = torch.sparse_coo_tensor(
+            torch.stack([self.inds[mode], col]), self.vals,
+            (self.dims[mode], ncols)).coalesce()
+        return coo.to_sparse_csr()
+
     # ----------------------------------------------------------- repair ops
 
     def fixed(self, dedup: bool = True, compress: bool = False) -> "SpTensor":

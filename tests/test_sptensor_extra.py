@@ -17,3 +17,17 @@ def test_zipf_synthetic_is_skewed():
     out = sp.mttkrp(cs, mats, 1)
     ref = sp.mttkrp_stream(t, mats, 1)
     assert (out - ref).abs().max() < 1e-9
+
+
+def test_unfold_matches_dense():
+    import torch
+    import splatt_amd as sp
+    t = sp.SpTensor.synthetic([7, 5, 6], 80, seed=3)
+    dense = torch.zeros(*t.dims, dtype=t.vals.dtype)
+    dense.index_put_((t.inds[0], t.inds[1], t.inds[2]), t.vals,
+                     accumulate=True)   # synthetic may contain duplicates
+    for mode in range(3):
+        X = t.unfold(mode).to_dense()
+        others = [m for m in range(3) if m != mode]
+        ref = dense.permute(mode, *others).reshape(t.dims[mode], -1)
+        assert torch.allclose(X, ref), mode
