@@ -129,6 +129,22 @@ def cmd_convert(args) -> int:
     elif kind == "hgraph":
         from splatt_amd.graph import hgraph_nnz, hgraph_write
         hgraph_write(hgraph_nnz(t), args.output)
+    elif kind == "fib_hgraph":
+        from splatt_amd.graph import hgraph_fib, hgraph_write
+        hgraph_write(hgraph_fib(t, args.mode), args.output)
+    elif kind == "csr":
+        # mode-`--mode` CSR unfolding (reference CNV_FIB_SPMAT,
+        # convert.c:134): "nrows ncols nnz" header then one
+        # "col val" pair list per row
+        X = t.unfold(args.mode)
+        cp = X.crow_indices().tolist()
+        ci = X.col_indices().tolist()
+        v = X.values().tolist()
+        with open(args.output, "w") as f:
+            f.write(f"{X.shape[0]} {X.shape[1]} {len(v)}\n")
+            for r in range(X.shape[0]):
+                f.write(" ".join(f"{ci[i] + 1} {v[i]:g}"
+                                 for i in range(cp[r], cp[r + 1])) + "\n")
     print(f"wrote {args.output} ({kind})")
     return 0
 
@@ -213,7 +229,10 @@ def main(argv=None) -> int:
     _add_common(p)
     p.add_argument("output")
     p.add_argument("-t", "--type", default="auto",
-                   choices=["auto", "bin", "tns", "graph", "hgraph"])
+                   choices=["auto", "bin", "tns", "graph", "hgraph",
+                            "fib_hgraph", "csr"])
+    p.add_argument("-m", "--mode", type=int, default=0,
+                   help="mode for fib_hgraph/csr conversions")
     p.set_defaults(fn=cmd_convert)
 
     p = sub.add_parser("stats", help="print tensor statistics")

@@ -224,3 +224,20 @@ def test_cli_cpd_deterministic_flag(tmp_path, capsys):
                "--deterministic", "--nowrite"])
     assert rc == 0
     assert os.environ.pop("SPLATT_DETERMINISTIC", None) == "1"
+
+
+def test_cli_convert_csr_and_fib_hgraph(tmp_path):
+    import splatt_amd as sp
+    from splatt_amd.cli import main
+    t = sp.SpTensor.synthetic([6, 5, 4], 40, seed=8).fixed()
+    f = tmp_path / "t.tns"
+    t.save(f)
+    out = tmp_path / "t.csr"
+    assert main(["convert", str(f), str(out), "-t", "csr", "-m", "1"]) == 0
+    lines = out.read_text().strip().splitlines()
+    nr, nc, nnz = map(int, lines[0].split())
+    assert (nr, nc) == (5, 24) and nnz == t.nnz
+    assert len(lines) == 1 + nr
+    out2 = tmp_path / "t.fib.hg"
+    assert main(["convert", str(f), str(out2), "-t", "fib_hgraph"]) == 0
+    assert out2.exists()
