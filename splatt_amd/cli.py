@@ -152,6 +152,11 @@ def cmd_convert(args) -> int:
 def cmd_stats(args) -> int:
     t = _load(args)
     print(stats_tt(t, args.tensor))
+    if getattr(args, "part", None):
+        from splatt_amd.graph import part_read
+        from splatt_amd.stats import stats_hparts
+        print(stats_hparts(t, part_read(args.part), args.part))
+        return 0
     cs = sp.csf_alloc(t, args.csf)
     print(stats_csf(cs))
     return 0
@@ -237,6 +242,8 @@ def main(argv=None) -> int:
 
     p = sub.add_parser("stats", help="print tensor statistics")
     _add_common(p)
+    p.add_argument("--part", help="nnz partition file: report partition "
+                                  "quality (reference --hparts analysis)")
     p.add_argument("--csf", default="two", choices=["one", "two", "all"])
     p.set_defaults(fn=cmd_stats)
 

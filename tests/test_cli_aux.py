@@ -241,3 +241,18 @@ def test_cli_convert_csr_and_fib_hgraph(tmp_path):
     out2 = tmp_path / "t.fib.hg"
     assert main(["convert", str(f), str(out2), "-t", "fib_hgraph"]) == 0
     assert out2.exists()
+
+
+def test_cli_stats_hparts(tmp_path, capsys):
+    import torch
+    import splatt_amd as sp
+    from splatt_amd.cli import main
+    t = sp.SpTensor.synthetic([10, 8, 12], 200, seed=9).fixed()
+    f = tmp_path / "t.tns"
+    t.save(f)
+    part = (torch.arange(t.nnz) * 3 // t.nnz)    # 3 contiguous parts
+    pf = tmp_path / "t.part"
+    pf.write_text("\n".join(str(int(x)) for x in part) + "\n")
+    assert main(["stats", str(f), "--part", str(pf)]) == 0
+    out = capsys.readouterr().out
+    assert "NPARTS=3" in out and "cut slices" in out
