@@ -51,9 +51,12 @@ def _mttkrp_ttbox(t, mats, mode):
 def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
                  niters: int = 3, device: str = "cpu",
                  policy: str = "all", validate: bool = False,
-                 seed: int = 123) -> Dict[str, dict]:
+                 seed: int = 123,
+                 threads: List[int] = None) -> Dict[str, dict]:
     """Time every algorithm for every mode; returns per-alg results with
-    seconds per mode and effective GFLOP/s (3*nnz*rank flops/MTTKRP)."""
+    seconds per mode and effective GFLOP/s (3*nnz*rank flops/MTTKRP).
+    `threads`: CPU thread counts to sweep for the CSF algorithms — the
+    reference bench's thread-scaling mode (bench.c --threads)."""
     algs = list(algs or ALGS)
     dev = torch.device(device)
     mats = [seeded_init(d, rank, m, seed).to(dev)
@@ -71,7 +74,11 @@ def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
 
     results: Dict[str, dict] = {}
     flops = 3.0 * t.nnz * rank
-    for alg in algs:
+    sweep = [(alg, nt) for alg in algs
+             for nt in ((threads or [0]) if (threads and dev.type == "cpu"
+                                             and alg in ("flat", "csf"))
+                        else [0])]
+    for alg, nt in sweep:
         per_mode = []
         ok = True
         for mode in range(t.nmodes):
@@ -82,7 +89,7 @@ def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
                     return _mttkrp_giga(td, mats, mode)
                 if alg == "ttbox":
                     return _mttkrp_ttbox(td, mats, mode)
-                return mttkrp(cs, mats, mode, alg=alg)
+                return mttkrp(cs, mats, mode, alg=alg, nthreads=nt)
             out = run()  # warmup + result for validation
             if gold is not None:
                 err = (out.cpu().double() - gold[mode].double()).abs().max()
@@ -93,7 +100,7 @@ def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
                 out = run()
             sync()
             per_mode.append((time.perf_counter() - tic) / niters)
-        results[alg] = {
+        results[alg if nt == 0 else f"{alg}@t{nt}"] = {
             "seconds_per_mode": per_mode,
             "gflops_per_mode": [flops / s / 1e9 for s in per_mode],
             "validated": ok if gold is not None else None,

@@ -169,8 +169,10 @@ def cmd_bench(args) -> int:
     dev = args.device
     if dev == "auto":
         dev = "cuda" if torch.cuda.is_available() else "cpu"
+    threads = [int(x) for x in args.threads.split(",")] if args.threads \
+        else None
     res = bench_mttkrp(t, args.rank, args.algs.split(","), args.iters,
-                       device=dev, validate=args.validate)
+                       device=dev, validate=args.validate, threads=threads)
     print(format_bench(res))
     return 0
 
@@ -254,6 +256,8 @@ def main(argv=None) -> int:
     p.add_argument("-N", "--iters", type=int, default=3)
     p.add_argument("--device", default="auto", choices=["auto", "cpu", "cuda"])
     p.add_argument("--validate", action="store_true")
+    p.add_argument("--threads", help="comma list of CPU thread counts to "
+                                     "sweep (reference bench scaling mode)")
     p.set_defaults(fn=cmd_bench)
 
     p = sub.add_parser("reorder", help="reorder a tensor")

@@ -256,3 +256,17 @@ def test_cli_stats_hparts(tmp_path, capsys):
     assert main(["stats", str(f), "--part", str(pf)]) == 0
     out = capsys.readouterr().out
     assert "NPARTS=3" in out and "cut slices" in out
+
+
+def test_cli_bench_thread_sweep(tmp_path, capsys):
+    import splatt_amd as sp
+    from splatt_amd.cli import main
+    t = sp.SpTensor.synthetic([15, 12, 18], 600, seed=2).fixed()
+    f = tmp_path / "t.tns"
+    t.save(f)
+    rc = main(["bench", str(f), "-r", "4", "-a", "flat,stream", "-N", "1",
+               "--device", "cpu", "--threads", "1,2", "--validate"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "flat@t1" in out and "flat@t2" in out and "stream" in out
+    assert "MISMATCH" not in out
