@@ -298,10 +298,12 @@ def _rsag_worker(rank, world, file_store, result_q):
             assert torch.allclose(out, ref), (n, rank)
 
         # end-to-end: grid CPD fit must equal the single-process fit.
-        # grid [2,1,2] at world 4: mode 1 has chunkn=30, layer size 4 ->
-        # per=8 with a 6-row last owner + 2 pad rows (uneven path).
+        # world 4, grid [2,1,2]: mode 1 has chunkn=30, layer size 4 ->
+        # per=8 with a 6-row last owner + 2 pad rows. world 3 (coarse
+        # [1,1,3]): 3 divides no mode dim -> every mode remainder path.
         t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
-        dec = GridDecomp.create(list(DIMS), grid=[2, 1, 2])
+        dec = GridDecomp.create(
+            list(DIMS), grid=[2, 1, 2] if world == 4 else [1, 1, 3])
         shard = dec.localize(t)
         cs = build_shard_csf(shard, list(DIMS), "two")
         opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
@@ -313,10 +315,13 @@ def _rsag_worker(rank, world, file_store, result_q):
 
 
 @pytest.mark.timeout(300)
-def test_grid_cpd_true_rs_primitives(tmp_path):
+@pytest.mark.parametrize("world", [3, 4])
+def test_grid_cpd_true_rs_primitives(tmp_path, world):
     """Force the RCCL reduce-scatter/all-gather tensor primitives (padded
-    equal chunks) under gloo and require the world-4 fit to equal the
-    single-process fit — covers the branch only nccl takes in production."""
+    equal chunks) under gloo and require the world-N fit to equal the
+    single-process fit — covers the branch only nccl takes in production.
+    world=3 is a non-divisor of every mode dim (the reference tests np=7
+    for the same remainder-path reason)."""
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     t = sp.SpTensor.synthetic(DIMS, NNZ, seed=SEED)
     opts = sp.CpdOptions(max_iters=ITERS, tolerance=0.0, seed=SEED)
@@ -324,9 +329,9 @@ def test_grid_cpd_true_rs_primitives(tmp_path):
 
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    store = str(tmp_path / "store_rsag")
-    procs = [ctx.Process(target=_rsag_worker, args=(r, 4, store, q))
-             for r in range(4)]
+    store = str(tmp_path / f"store_rsag{world}")
+    procs = [ctx.Process(target=_rsag_worker, args=(r, world, store, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     tag, fit4 = q.get()
