@@ -136,3 +136,23 @@ def test_cli_binary_missing_file():
                        capture_output=True, text=True)
     assert r.returncode == 1
     assert "cannot open" in r.stderr
+
+
+def test_capi_example_compiles_and_runs(tmp_path):
+    """examples/capi_demo.c must build against csrc/capi/splatt.h and run
+    end to end (load -> CPD -> MTTKRP) on a small tensor."""
+    t = sp.SpTensor.synthetic([20, 15, 25], 800, seed=6).fixed()
+    tns = tmp_path / "t.tns"
+    t.save(tns)
+    exe = tmp_path / "capi_demo"
+    r = subprocess.run(
+        ["gcc", "-O2", os.path.join(ROOT, "examples", "capi_demo.c"),
+         "-I" + os.path.join(ROOT, "csrc", "capi"),
+         "-L" + os.path.join(ROOT, "bin"), "-lsplatt",
+         "-Wl,-rpath," + os.path.join(ROOT, "bin"), "-o", str(exe)],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-500:]
+    r = subprocess.run([str(exe), str(tns), "6"], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0, (r.stdout, r.stderr[-300:])
+    assert "CPD fit:" in r.stdout and "mttkrp mode 0" in r.stdout
