@@ -18,18 +18,25 @@ FilePtr xopen(const std::string & path, const char * mode) {
   return FilePtr(f);
 }
 
-// parse one line of whitespace-separated numbers; returns token count
-inline int parse_line(const char * s, double * out, int maxtok) {
+// parse one line of whitespace-separated numbers; returns token count.
+// `clean` (optional) reports whether the whole line was consumed — false
+// means a non-numeric token or more tokens than maxtok (malformed input,
+// which a silent parser would truncate into wrong coordinates).
+inline int parse_line(const char * s, double * out, int maxtok,
+                      bool * clean = nullptr) {
   int n = 0;
-  while (*s && n < maxtok) {
+  bool ok = true;
+  while (*s) {
     while (*s && std::isspace((unsigned char)*s)) ++s;
     if (!*s || *s == '#' || *s == '%') break;
+    if (n >= maxtok) { ok = false; break; }
     char * end = nullptr;
     out[n] = strtod(s, &end);
-    if (end == s) break;
+    if (end == s) { ok = false; break; }
     s = end;
     ++n;
   }
+  if (clean) *clean = ok;
   return n;
 }
 
@@ -48,9 +55,17 @@ SpTensor<V> tns_read(const std::string & path) {
   idx_t dims[MAX_NMODES] = {0};
   idx_t minidx = ~(idx_t)0;
   ssize_t r;
+  idx_t lineno = 0;
   while ((r = getline(&line, &cap, f.get())) >= 0) {
-    const int n = parse_line(line, tok, MAX_NMODES + 2);
-    if (n < 2) continue;
+    ++lineno;
+    bool clean = true;
+    const int n = parse_line(line, tok, MAX_NMODES + 2, &clean);
+    if (n == 0 && clean) continue;          // blank / comment line
+    if (!clean || n < 2 || (nmodes != 0 && n != nmodes + 1)) {
+      free(line);
+      throw std::runtime_error("malformed line " + std::to_string(lineno) +
+                               " in " + path);
+    }
     if (nmodes == 0) {
       nmodes = n - 1;
       if (nmodes < 1 || nmodes > MAX_NMODES) {
@@ -64,6 +79,10 @@ SpTensor<V> tns_read(const std::string & path) {
       minidx = std::min(minidx, v);
     }
     ++nnz;
+  }
+  if (nnz == 0) {
+    free(line);
+    throw std::runtime_error("no nonzeros found in " + path);
   }
   const idx_t offset = (minidx == 0) ? 0 : 1;  // 0- vs 1-indexed autodetect
   for (int m = 0; m < nmodes; ++m) dims[m] += 1 - offset;

@@ -1,5 +1,6 @@
 """I/O tests (reference tests/io_test.c): text round-trip, 0/1-index
 equivalence, binary round-trip incl. width down-conversion."""
+import pytest
 import torch
 
 import splatt_amd as sp
@@ -91,3 +92,25 @@ def test_huge_dim_rejected():
                     [2, 2**33])
     with pytest.raises((ValueError, RuntimeError)):
         build_csf(t, [0, 1])
+
+
+@pytest.mark.parametrize("content,msg", [
+    ("1 2 x 3.0\n", "malformed line 1"),
+    ("", "no nonzeros"),
+    ("1 2 3 4.0\n1 2 3\n", "malformed line 2"),          # short line
+    ("1 2 3 4.0\n1 2 3 4.0 5.0\n", "malformed line 2"),  # extra token
+])
+def test_tns_rejects_malformed(tmp_path, content, msg):
+    """Garbage tokens / inconsistent arity / empty files must raise, not
+    silently truncate into wrong coordinates."""
+    f = tmp_path / "bad.tns"
+    f.write_text(content)
+    with pytest.raises(RuntimeError, match=msg):
+        sp.load(str(f))
+
+
+def test_tns_accepts_comments_and_blanks(tmp_path):
+    f = tmp_path / "ok.tns"
+    f.write_text("# header comment\n\n1 2 3 4.0\n% other comment\n2 3 1 5.0\n")
+    t = sp.load(str(f))
+    assert t.nnz == 2 and t.dims == [2, 3, 3]
