@@ -66,7 +66,11 @@ class SpTensor:
         coo = torch.sparse_coo_tensor(
             torch.stack([self.inds[mode], col]), self.vals,
             (self.dims[mode], ncols)).coalesce()
-        return coo.to_sparse_csr()
+        import warnings
+        with warnings.catch_warnings():
+            # torch's blanket "CSR support is in beta" notice
+            warnings.simplefilter("ignore", UserWarning)
+            return coo.to_sparse_csr()
 
     # ----------------------------------------------------------- repair ops
 
