@@ -1,5 +1,5 @@
-"""Command-line interface: `python -m splatt_amd <cmd>` (or the `splatt`
-console script).
+"""Command-line interface: `python -m splatt_amd <cmd>` (the native
+`bin/splatt` binary covers the host-library surface).
 
 Capability parity: the reference `splatt` binary and its sub-commands
 (cmds/splatt_cmds.h:18-27): cpd, bench, check, convert, reorder, stats.
