@@ -46,6 +46,17 @@ class GraphStepRunner:
         self.fit_parts = torch.zeros(2, dtype=torch.float64, device=dev)
         self.buf = st.buf
         self.tmp = torch.empty_like(st.buf)
+        # deterministic mode: static workspaces for the atomic-free
+        # solve/gram kernels (mttkrp already reads the env per call)
+        import os
+        self.det = (os.environ.get("SPLATT_DETERMINISTIC") == "1"
+                    and F in (4, 8, 16, 32, 64))
+        if self.det:
+            self.gpart = []
+            for m in range(nm):
+                nparts = max(1, min(256, (dec.chunkn[m] + 63) // 64))
+                self.gpart.append(torch.empty(nparts * F * F, dtype=dtype,
+                                              device=dev))
 
     def _static_step(self):
         st, nm = self.st, self.nm
@@ -66,13 +77,20 @@ class GraphStepRunner:
                 else:
                     self.G.mul_(self.grams[o])
             native().gpu_spd_inverse(self.G, self.Ginv, stream)
-            torch.mm(mb, self.Ginv, out=self.A[m])
+            if self.det:
+                native().gpu_rowsolve(mb, self.Ginv, self.A[m], stream)
+            else:
+                torch.mm(mb, self.Ginv, out=self.A[m])
             # max-norm normalize (steady state)
             torch.amax(torch.abs(self.A[m]), dim=0, out=self.lam)
             self.lam.clamp_(min=1.0)
             self.A[m].div_(self.lam)
-            self.grams[m].zero_()
-            native().gpu_gram(self.A[m], self.grams[m], stream)
+            if self.det:
+                native().gpu_gram_det(self.A[m], self.gpart[m],
+                                      self.grams[m], stream)
+            else:
+                self.grams[m].zero_()
+                native().gpu_gram(self.A[m], self.grams[m], stream)
         # fit parts on device: [inner, knorm]
         mlast = nm - 1
         n = dec.chunkn[mlast]
@@ -88,7 +106,8 @@ class GraphStepRunner:
             else:
                 self.G.mul_(self.grams[o])
         lamd = self.lam.double()
-        self.fit_parts[1] = lamd @ self.G.double() @ lamd
+        self.fit_parts[1] = (self.G.double()
+                             * torch.outer(lamd, lamd)).sum()
 
     def capture(self) -> bool:
         self.capture_error = None

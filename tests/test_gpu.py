@@ -401,3 +401,38 @@ def test_gpu_gram_det_matches():
         del os.environ["SPLATT_DETERMINISTIC"]
     assert (G - A.T @ A).abs().max().item() < 1e-8
     assert torch.equal(G, G2)
+
+
+def test_gpu_deterministic_graph_replay():
+    """hipGraph-captured iteration under SPLATT_DETERMINISTIC=1: two
+    independent capture+replay sequences must produce bitwise-equal
+    factors (the graph path routes solve/gram through the det kernels)."""
+    import os
+    from splatt_amd.parallel.grid import GridDecomp, grid_cpd_init, \
+        grid_cpd_step
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    from splatt_amd.parallel.graph_exec import GraphStepRunner
+
+    def run():
+        t = sp.SpTensor.synthetic([300, 260, 420], 150_000, seed=71,
+                                  device="cuda")
+        dec = GridDecomp.create([300, 260, 420])
+        cs = build_shard_csf(t, [300, 260, 420], "all", flat_only=True)
+        st = grid_cpd_init(cs, dec, 16, sp.CpdOptions(seed=5))
+        grid_cpd_step(st, 0)   # iteration 0 eager (2-norm schedule)
+        r = GraphStepRunner(st)
+        assert r.capture(), r.capture_error
+        for _ in range(3):
+            r.replay()
+        r.finalize(st.norm_x)
+        return [a.clone() for a in st.factors], st.fit
+
+    os.environ["SPLATT_DETERMINISTIC"] = "1"
+    try:
+        f1, fit1 = run()
+        f2, fit2 = run()
+    finally:
+        del os.environ["SPLATT_DETERMINISTIC"]
+    assert fit1 == fit2
+    for a, b in zip(f1, f2):
+        assert torch.equal(a, b)
