@@ -6,6 +6,8 @@
 #include <cstdio>
 #include <cstring>
 #include <string>
+#include <array>
+#include <vector>
 #include <chrono>
 
 #include "splatt.h"
@@ -25,7 +27,9 @@ static void usage() {
       "  splatt check TENSOR [--fix OUT]\n"
       "  splatt convert TENSOR OUT          (.bin <-> .tns by extension)\n"
       "  splatt stats TENSOR\n"
-      "  splatt bench TENSOR [-r RANK] [-N ITERS]\n");
+      "  splatt bench TENSOR [-r RANK] [-N ITERS]\n"
+      "  splatt reorder TENSOR OUT [--type rand|perm] [--seed S]"
+      " [--permfile PREFIX]\n");
 }
 
 static double argf(int argc, char ** argv, const char * flag, double dflt) {
@@ -124,6 +128,73 @@ static int cmd_stats(int argc, char ** argv) {
   return 0;
 }
 
+static int cmd_reorder(int argc, char ** argv) {
+  // splatt reorder TENSOR OUT [--type rand|perm] [--seed S]
+  //                [--permfile PREFIX]
+  // rand: fresh random permutation per mode; perm: read PREFIX.modeM.perm
+  // (the Python CLI's file format; it also offers graph/hgraph-driven
+  // orders). --permfile with rand writes the generated permutation.
+  if (argc < 2) { usage(); return 1; }
+  auto tt = tensor_load<double>(argv[0]);
+  const char * type = args(argc, argv, "--type");
+  const char * pfx = args(argc, argv, "--permfile");
+  const uint64_t seed = (uint64_t)argf(argc, argv, "--seed", 42);
+  const bool apply_file = type && !std::strcmp(type, "perm");
+  if (apply_file && !pfx) {
+    std::fprintf(stderr, "reorder --type perm needs --permfile\n");
+    return 1;
+  }
+  std::array<std::vector<idx_t>, MAX_NMODES> perm;
+  for (int m = 0; m < tt.nmodes; ++m) {
+    perm[m].resize(tt.dims[m]);
+    if (apply_file) {
+      char name[512];
+      std::snprintf(name, sizeof name, "%s.mode%d.perm", pfx, m);
+      FILE * f = std::fopen(name, "r");
+      if (!f) throw std::runtime_error(std::string("cannot open ") + name);
+      unsigned long long v;
+      for (idx_t i = 0; i < tt.dims[m]; ++i) {
+        if (std::fscanf(f, "%llu", &v) != 1 || v >= tt.dims[m]) {
+          std::fclose(f);
+          throw std::runtime_error(std::string("bad perm file ") + name);
+        }
+        perm[m][i] = (idx_t)v;
+      }
+      std::fclose(f);
+    } else {
+      for (idx_t i = 0; i < tt.dims[m]; ++i) perm[m][i] = i;
+      // Fisher-Yates with splitmix64-style mixing per mode
+      uint64_t s = seed * 0x9E3779B97F4A7C15ull + (uint64_t)m + 1;
+      for (idx_t i = tt.dims[m] - 1; i > 0; --i) {
+        s += 0x9E3779B97F4A7C15ull;
+        uint64_t z = s;
+        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+        z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+        z ^= z >> 31;
+        std::swap(perm[m][i], perm[m][z % (i + 1)]);
+      }
+    }
+  }
+  for (int m = 0; m < tt.nmodes; ++m)
+    for (idx_t i = 0; i < tt.nnz; ++i)
+      tt.ind[m][i] = perm[m][tt.ind[m][i]];
+  tns_write(tt, argv[1]);
+  std::printf("wrote %s\n", argv[1]);
+  if (pfx && !apply_file) {
+    for (int m = 0; m < tt.nmodes; ++m) {
+      char name[512];
+      std::snprintf(name, sizeof name, "%s.mode%d.perm", pfx, m);
+      FILE * f = std::fopen(name, "w");
+      if (!f) throw std::runtime_error(std::string("cannot write ") + name);
+      for (idx_t i = 0; i < tt.dims[m]; ++i)
+        std::fprintf(f, "%llu\n", (unsigned long long)perm[m][i]);
+      std::fclose(f);
+    }
+    std::printf("wrote %s.mode*.perm\n", pfx);
+  }
+  return 0;
+}
+
 static int cmd_bench(int argc, char ** argv) {
   auto tt = tensor_load<double>(argv[0]);
   print_stats(tt, argv[0]);
@@ -166,6 +237,7 @@ int main(int argc, char ** argv) {
     if (cmd == "convert") return cmd_convert(sub_argc, sub_argv);
     if (cmd == "stats") return cmd_stats(sub_argc, sub_argv);
     if (cmd == "bench") return cmd_bench(sub_argc, sub_argv);
+    if (cmd == "reorder") return cmd_reorder(sub_argc, sub_argv);
   } catch (const std::exception & e) {
     std::fprintf(stderr, "splatt: %s\n", e.what());
     return 1;

@@ -156,3 +156,28 @@ def test_capi_example_compiles_and_runs(tmp_path):
                        text=True, timeout=120)
     assert r.returncode == 0, (r.stdout, r.stderr[-300:])
     assert "CPD fit:" in r.stdout and "mttkrp mode 0" in r.stdout
+
+
+def test_cli_binary_reorder(tmp_path):
+    """Native `splatt reorder`: random relabeling preserves shape/values;
+    --type perm re-applies a written permutation reproducibly."""
+    t = sp.SpTensor.synthetic([12, 9, 15], 300, seed=4).fixed()
+    tns = tmp_path / "t.tns"
+    t.save(tns)
+    out1 = tmp_path / "r1.tns"
+    out2 = tmp_path / "r2.tns"
+    pfx = str(tmp_path / "p")
+    r = subprocess.run([EXE, "reorder", str(tns), str(out1), "--seed", "9",
+                        "--permfile", pfx], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    r = subprocess.run([EXE, "reorder", str(tns), str(out2), "--type",
+                        "perm", "--permfile", pfx],
+                       capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    assert out1.read_text() == out2.read_text()
+    t1 = sp.load(str(out1))
+    assert t1.dims == t.dims and t1.nnz == t.nnz
+    assert torch.allclose(t1.vals.sort().values, t.vals.sort().values)
+    for m in range(3):
+        assert torch.equal(torch.bincount(t1.inds[m]).sort().values,
+                           torch.bincount(t.inds[m]).sort().values)
