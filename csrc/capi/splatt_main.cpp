@@ -1,8 +1,9 @@
 // `splatt` CLI binary (native, torch-free).
 // Parity: reference cmds/splatt_bin.c:76-123 dispatcher with sub-commands
-// cpd / check / convert / stats / bench (cmds/cmd_*.c). The Python CLI
-// (`python -m splatt_amd`) is the full-featured front end (GPU paths,
-// reorder, graph exports); this binary covers the host library surface.
+// cpd / check / convert / stats / bench / reorder (cmds/cmd_*.c). The
+// Python CLI (`python -m splatt_amd`) is the full-featured front end
+// (GPU paths, graph exports, graph/hgraph-driven reorders); this binary
+// covers the host library surface.
 #include <cstdio>
 #include <cstring>
 #include <string>
