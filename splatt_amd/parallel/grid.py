@@ -407,11 +407,12 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
     rs_part = getattr(st, "_rs_inner", None)
     if rs_part is not None:
         _ar(rs_part)
-        inner_t = rs_part @ st.lam.double()
+        inner_t = (rs_part * st.lam.double()).sum()
         st._rs_inner = None  # type: ignore[attr-defined]
     else:
-        inner_t = (st.buf[: dec.chunkn[mlast]].double()
-                   * st.factors[mlast].double()).sum(dim=0) @ st.lam.double()
+        inner_t = ((st.buf[: dec.chunkn[mlast]].double()
+                    * st.factors[mlast].double()).sum(dim=0)
+                   * st.lam.double()).sum()
         if dec.grid[mlast] > 1:
             inner_t /= dec.repl(mlast)
             _ar(inner_t)
@@ -419,7 +420,8 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
     Gall = torch.ones(F, F, dtype=dtype, device=dev)
     for o in range(nm):
         Gall *= st.grams[o]
-    knorm = float(st.lam.double() @ Gall.double() @ st.lam.double())
+    knorm = float((Gall.double()
+                   * torch.outer(st.lam.double(), st.lam.double())).sum())
     residual = math.sqrt(max(0.0, st.norm_x + knorm - 2 * inner))
     st.old_fit = st.fit
     st.fit = 1.0 - residual / math.sqrt(st.norm_x)
