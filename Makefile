@@ -1,5 +1,5 @@
 # Convenience targets (the canonical build is setup.py / __graft_entry__.build)
-.PHONY: build test gpu-test bench clean
+.PHONY: build test gpu-test bench asan soak clean
 
 build:
 	python setup.py build_ext --inplace
@@ -12,6 +12,12 @@ gpu-test: build
 
 bench: build
 	python bench.py --steps 10 --warmup 3
+
+asan:
+	bash scripts/asan.sh
+
+soak:
+	python scripts/soak.py
 
 clean:
 	rm -rf build bin splatt_amd/_C*.so splatt_amd/__pycache__ .pytest_cache
