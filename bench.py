@@ -82,11 +82,16 @@ def main():
                     help="-1 (default): LDS-staged bucketing sized by "
                          "SPLATT_LDS_KB; 0: off; N>1: plain gather-range "
                          "buckets without LDS staging")
+    ap.add_argument("--deterministic", action="store_true",
+                    help="bitwise-reproducible kernels (SPLATT_DETERMINISTIC=1;"
+                         " ~60%% of default throughput)")
     ap.add_argument("--decomp", default="coarse", choices=["coarse", "medium"],
                     help="coarse = 1D layers on the longest mode (weak "
                          "scaling); medium = nmodes-D grid (strong scaling)")
     args = ap.parse_args()
 
+    if args.deterministic:
+        os.environ["SPLATT_DETERMINISTIC"] = "1"
     dims, nnz_shard, rank_f, _ = CONFIGS[args.config]
     if args.rank_f:
         rank_f = args.rank_f
