@@ -27,6 +27,7 @@ static Options opts_from_array(const double * o) {
   Options opt;
   if (!o) return opt;
   opt.tolerance = o[SPLATT_OPTION_TOLERANCE];
+  opt.regularize = o[SPLATT_OPTION_REGULARIZE];
   opt.max_iters = (idx_t)o[SPLATT_OPTION_NITER];
   opt.verbosity = (int)o[SPLATT_OPTION_VERBOSITY];
   opt.nthreads = (int)o[SPLATT_OPTION_NTHREADS];
@@ -49,6 +50,12 @@ double * splatt_default_opts(void) {
   o[SPLATT_OPTION_NTHREADS] = 0;
   o[SPLATT_OPTION_RANDSEED] = (double)0x5EED5EEDull;
   o[SPLATT_OPTION_CSF_ALLOC] = SPLATT_CSF_TWOMODE;
+  o[SPLATT_OPTION_REGULARIZE] = 0.0;
+  o[SPLATT_OPTION_TILE] = 0.0;
+  o[SPLATT_OPTION_TILELEVEL] = 1.0;
+  o[SPLATT_OPTION_PRIVTHRESH] = 0.02;
+  o[SPLATT_OPTION_DECOMP] = 0.0;
+  o[SPLATT_OPTION_COMM] = 0.0;
   return o;
 }
 

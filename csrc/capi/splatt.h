@@ -53,6 +53,12 @@ typedef enum splatt_option_type {
   SPLATT_OPTION_NTHREADS,       /* OpenMP threads (omp default)       */
   SPLATT_OPTION_RANDSEED,       /* RNG seed (fixed default)           */
   SPLATT_OPTION_CSF_ALLOC,      /* splatt_csf_type (TWOMODE)          */
+  SPLATT_OPTION_REGULARIZE,     /* ridge term on the Gram diagonal (0)  */
+  SPLATT_OPTION_TILE,           /* advisory: gather-range bucketing     */
+  SPLATT_OPTION_TILELEVEL,      /* advisory (GPU build picks LDS tiles) */
+  SPLATT_OPTION_PRIVTHRESH,     /* advisory (GPU folds runs in regs)    */
+  SPLATT_OPTION_DECOMP,         /* advisory (Python layer selects)      */
+  SPLATT_OPTION_COMM,           /* advisory (RCCL collectives)          */
   SPLATT_OPTION_NOPTIONS
 } splatt_option_type;
 

@@ -24,7 +24,8 @@ using namespace splatt;
 static void usage() {
   std::printf(
       "splatt — MI355X-native sparse tensor factorization (host CLI)\n\n"
-      "  splatt cpd TENSOR [-r RANK] [-i ITERS] [-t TOL] [--seed S] [--nowrite]\n"
+      "  splatt cpd TENSOR [-r RANK] [-i ITERS] [-t TOL] [--seed S]"
+      " [--reg R] [--nowrite]\n"
       "  splatt check TENSOR [--fix OUT]\n"
       "  splatt convert TENSOR OUT          (.bin <-> .tns by extension)\n"
       "  splatt stats TENSOR\n"
@@ -66,6 +67,7 @@ static int cmd_cpd(int argc, char ** argv) {
   o[SPLATT_OPTION_NITER] = argf(argc, argv, "-i", 50);
   o[SPLATT_OPTION_TOLERANCE] = argf(argc, argv, "-t", 1e-5);
   o[SPLATT_OPTION_RANDSEED] = argf(argc, argv, "--seed", (double)0x5EED5EEDull);
+  o[SPLATT_OPTION_REGULARIZE] = argf(argc, argv, "--reg", 0.0);
   const int rank = (int)argf(argc, argv, "-r", 10);
 
   splatt_idx_t nmodes = 0;

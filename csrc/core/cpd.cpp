@@ -71,7 +71,8 @@ Kruskal<V> cpd_als(const CsfSet<V> & set, int rank, const Options & opts,
       std::memcpy(k.factors[m].data(), mttkrp_buf.data(),
                   sizeof(V) * c0.dims[m] * F);
       gram_hadamard(gram_ptrs.data(), nm, m, F, G.data());
-      solve_normals(k.factors[m].data(), c0.dims[m], F, G.data());
+      solve_normals(k.factors[m].data(), c0.dims[m], F, G.data(),
+                    (V)opts.regularize);
       // 2-norm on first iteration, max-norm after (reference cpd.c:343-347)
       mat_normalize(k.factors[m].data(), c0.dims[m], F, k.lambda.data(),
                     it == 0 ? 0 : 1);

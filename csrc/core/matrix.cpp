@@ -79,10 +79,10 @@ bool cholesky(V * L, int F) {
 }  // namespace
 
 template <typename V>
-int solve_normals(V * B, idx_t n, int F, const V * G) {
+int solve_normals(V * B, idx_t n, int F, const V * G, V reg0) {
   std::vector<V> L((size_t)F * F);
   int bumps = 0;
-  V reg = (V)0;
+  V reg = reg0;  // user ridge term (reference p_form_gram + reg*I)
   // escalate Tikhonov regularization until the Cholesky succeeds
   for (;;) {
     std::memcpy(L.data(), G, sizeof(V) * F * F);
@@ -158,7 +158,7 @@ void mat_normalize(V * A, idx_t n, int F, V * lambda, int which) {
 #define INST(V) \
   template void mat_ata<V>(const V*, idx_t, int, V*); \
   template void gram_hadamard<V>(V const* const*, int, int, int, V*, V); \
-  template int solve_normals<V>(V*, idx_t, int, const V*); \
+  template int solve_normals<V>(V*, idx_t, int, const V*, V); \
   template void mat_normalize<V>(V*, idx_t, int, V*, int);
 INST(float)
 INST(double)

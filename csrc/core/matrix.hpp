@@ -24,7 +24,8 @@ void gram_hadamard(V const * const * grams, int nmats, int skip, int F,
 // (the reference falls back to gelss SVD — we use Tikhonov instead and
 // report it via the return value: 0 = clean, k = #bumps applied).
 template <typename V>
-int solve_normals(V * B, idx_t n, int F, const V * G);
+int solve_normals(V * B, idx_t n, int F, const V * G,
+                  V reg0 = (V)0);
 
 // Column 2-norms (or max-norms) of A (n x F) into lambda, then scale columns
 // to unit norm. which: 0 = 2-norm, 1 = max-norm.

@@ -37,6 +37,7 @@ struct Options {
   TileMode tile      = TileMode::NOTILE;
   idx_t  tile_depth  = 1;       // number of leaf-side levels tiled
   double privatize_threshold = 0.02;
+  double regularize  = 0.0;     // ridge term added to the Gram diagonal
   uint64_t seed      = 0;       // 0 -> nondeterministic
   int    nthreads    = 0;       // 0 -> omp default
   int    verbosity   = 1;

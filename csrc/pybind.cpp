@@ -27,6 +27,8 @@ static sp::Options opts_from_dict(const py::dict & d) {
   if (d.contains("max_iters")) o.max_iters = d["max_iters"].cast<uint64_t>();
   if (d.contains("seed")) o.seed = d["seed"].cast<uint64_t>();
   if (d.contains("nthreads")) o.nthreads = d["nthreads"].cast<int>();
+  if (d.contains("regularize"))
+    o.regularize = d["regularize"].cast<double>();
   if (d.contains("csf_alloc")) {
     const std::string a = d["csf_alloc"].cast<std::string>();
     o.csf_alloc = a == "one" ? sp::CsfAlloc::ONEMODE

@@ -42,7 +42,8 @@ def cmd_cpd(args) -> int:
     print(stats_tt(t, args.tensor))
     opts = sp.CpdOptions(tolerance=args.tol, max_iters=args.its,
                          seed=args.seed, csf_alloc=args.csf,
-                         nthreads=args.nthreads, verbose=args.verbose > 0)
+                         nthreads=args.nthreads, verbose=args.verbose > 0,
+                         regularize=args.reg)
     dev = args.device
     if dev == "auto":
         dev = "cuda" if torch.cuda.is_available() else "cpu"
@@ -213,6 +214,8 @@ def main(argv=None) -> int:
     p.add_argument("-t", "--tol", type=float, default=1e-5)
     p.add_argument("-i", "--its", type=int, default=50)
     p.add_argument("--seed", type=int, default=0x5EED5EED)
+    p.add_argument("--reg", type=float, default=0.0,
+                   help="ridge regularization on the normal equations")
     p.add_argument("--csf", default="two", choices=["one", "two", "all"])
     p.add_argument("--device", default="auto", choices=["auto", "cpu", "cuda"])
     p.add_argument("--nthreads", type=int, default=0)

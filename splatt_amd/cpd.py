@@ -34,6 +34,8 @@ class CpdOptions:
     csf_alloc: str = "two"
     nthreads: int = 0
     verbose: bool = False
+    regularize: float = 0.0   # ridge term on the Gram diagonal
+                              # (reference p_form_gram + reg*I)
     # per-iteration factor checkpointing (the reference has none —
     # SURVEY.md §5 flags this as a cheap rebuild improvement)
     checkpoint_path: str = ""
@@ -113,6 +115,9 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             for o in range(nm):
                 if o != m:
                     G *= grams[o]
+            if opts.regularize:
+                G += opts.regularize * torch.eye(rank, dtype=dtype,
+                                                 device=dev)
             # solve A * G = mttkrp  =>  G^T A^T = mttkrp^T (G symmetric)
             # F x F inverse once, then one well-shaped (n x F)(F x F) GEMM —
             # beats a trsm against an n-row RHS at these tiny F
@@ -177,6 +182,7 @@ def cpd_als_cpu_native(t: SpTensor, rank: int,
             "seed": opts.seed,
             "csf_alloc": opts.csf_alloc,
             "nthreads": opts.nthreads,
+            "regularize": opts.regularize,
         })
     return Kruskal(factors=list(r["factors"]), lam=r["lambda"],
                    fit=float(r["fit"]), niters=int(r["niters"]))

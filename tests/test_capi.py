@@ -181,3 +181,18 @@ def test_cli_binary_reorder(tmp_path):
     for m in range(3):
         assert torch.equal(torch.bincount(t1.inds[m]).sort().values,
                            torch.bincount(t.inds[m]).sort().values)
+
+
+def test_capi_regularize_option(lib, tmp_path):
+    """SPLATT_OPTION_REGULARIZE flows into the C API CPD."""
+    t = sp.SpTensor.synthetic([20, 18, 22], 900, seed=15).fixed()
+    tns = tmp_path / "t.tns"
+    t.save(tns)
+    out = subprocess.run([EXE, "cpd", str(tns), "-r", "5", "-i", "6",
+                          "--nowrite"], capture_output=True, text=True)
+    base = float(out.stdout.split("Final fit:")[1].split()[0])
+    out = subprocess.run([EXE, "cpd", str(tns), "-r", "5", "-i", "6",
+                          "--reg", "25.0", "--nowrite"],
+                         capture_output=True, text=True)
+    reg = float(out.stdout.split("Final fit:")[1].split()[0])
+    assert reg < base

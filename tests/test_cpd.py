@@ -92,3 +92,23 @@ def test_checkpoint_every_interval(tmp_path, small3):
     # last multiple-of-2 iteration is it=3 (0-indexed), i.e. 4 iterations
     assert state["iteration"] == 3
     assert len(state["factors"]) == 3
+
+
+def test_cpd_regularize():
+    """Ridge term: reg=0 reproduces the unregularized run exactly; a
+    large reg shrinks the solution (lower fit) but still converges
+    (reference p_form_gram's + reg*I, matrix.c:29-83)."""
+    import splatt_amd as sp
+    t = sp.SpTensor.synthetic([25, 20, 30], 2000, seed=13)
+    o0 = sp.CpdOptions(max_iters=8, tolerance=0.0, seed=3)
+    oz = sp.CpdOptions(max_iters=8, tolerance=0.0, seed=3, regularize=0.0)
+    ob = sp.CpdOptions(max_iters=8, tolerance=0.0, seed=3, regularize=10.0)
+    k0 = sp.cpd_als(t, 6, o0)
+    kz = sp.cpd_als(t, 6, oz)
+    kb = sp.cpd_als(t, 6, ob)
+    assert k0.fit == kz.fit
+    assert kb.fit < k0.fit
+    assert kb.fit == kb.fit  # finite
+    # host reference path honors it too
+    kn = sp.cpd_als_cpu_native(t, 6, ob)
+    assert kn.fit < sp.cpd_als_cpu_native(t, 6, o0).fit + 1e-12
