@@ -436,3 +436,12 @@ def test_gpu_deterministic_graph_replay():
     assert fit1 == fit2
     for a, b in zip(f1, f2):
         assert torch.equal(a, b)
+
+
+def test_gpu_cpd_regularize():
+    t = sp.SpTensor.synthetic([100, 80, 120], 30_000, seed=19)
+    o0 = sp.CpdOptions(max_iters=5, tolerance=0.0, seed=3)
+    ob = sp.CpdOptions(max_iters=5, tolerance=0.0, seed=3, regularize=20.0)
+    k0 = sp.cpd_als(sp.csf_alloc(t.to("cuda"), "all"), 8, o0)
+    kb = sp.cpd_als(sp.csf_alloc(t.to("cuda"), "all"), 8, ob)
+    assert kb.fit < k0.fit and kb.fit == kb.fit
