@@ -26,3 +26,27 @@ def test_bench_torchrun_world2(decomp, tmp_path):
     j = json.loads(lines[0])
     assert j["n_gpus"] == 2
     assert j["scaling"] == ("weak" if decomp == "coarse" else "strong")
+
+
+@pytest.mark.timeout(420)
+def test_cli_cpd_torchrun_world2(tmp_path):
+    """`torchrun -m splatt_amd cpd` — the mpirun-splatt-cpd analog — must
+    run end to end on CPU/gloo and write rank-0 factor files."""
+    import torch  # noqa: F401  (ensures torch importable before subprocess)
+    import splatt_amd as sp
+    t = sp.SpTensor.synthetic([60, 45, 80], 4000, seed=21).fixed()
+    tns = tmp_path / "t.tns"
+    t.save(tns)
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               PYTHONPATH=ROOT + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--nnodes=1", "--nproc-per-node", "2", "--local-addr", "127.0.0.1",
+         "-m", "splatt_amd", "cpd", str(tns), "-r", "6", "-i", "4", "-t", "0",
+         "--device", "cpu"],
+        capture_output=True, text=True, cwd=str(tmp_path), env=env,
+        timeout=400)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "Final fit:" in r.stdout
+    for m in range(3):
+        assert (tmp_path / f"mode{m + 1}.mat").exists()
