@@ -54,8 +54,8 @@ double * splatt_default_opts(void) {
   o[SPLATT_OPTION_TILE] = 0.0;
   o[SPLATT_OPTION_TILELEVEL] = 1.0;
   o[SPLATT_OPTION_PRIVTHRESH] = 0.02;
-  o[SPLATT_OPTION_DECOMP] = 0.0;
-  o[SPLATT_OPTION_COMM] = 0.0;
+  o[SPLATT_OPTION_DECOMP] = SPLATT_DECOMP_MEDIUM;
+  o[SPLATT_OPTION_COMM] = SPLATT_COMM_ALL2ALL;
   return o;
 }
 

@@ -27,10 +27,15 @@ typedef double splatt_val_t;
 
 #define SPLATT_MAX_NMODES 8
 
+/* Numeric values below match the reference public header exactly
+ * (reference include/splatt/types_config.h:103-215) so clients compiled
+ * against either header agree on option slots and return codes:
+ * options NTHREADS=0, TOLERANCE=1, REGULARIZE=2, NITER=3, VERBOSITY=4,
+ * then RANDSEED..COMM; SPLATT_SUCCESS=1 with positive error codes. */
 typedef enum splatt_error_type {
-  SPLATT_SUCCESS = 0,
-  SPLATT_ERROR_BADINPUT = -1,
-  SPLATT_ERROR_NOMEMORY = -2,
+  SPLATT_SUCCESS = 1,
+  SPLATT_ERROR_BADINPUT = 2,
+  SPLATT_ERROR_NOMEMORY = 3,
 } splatt_error_type;
 
 typedef enum splatt_verbosity_type {
@@ -46,19 +51,38 @@ typedef enum splatt_csf_type {
   SPLATT_CSF_ALLMODE = 2,
 } splatt_csf_type;
 
+typedef enum splatt_tile_type {
+  SPLATT_NOTILE = 0,
+  SPLATT_DENSETILE = 1,
+  SPLATT_SYNCTILE = 2,   /* deprecated in the reference; accepted, mapped
+                            to the gather-range bucketing analog */
+  SPLATT_COOPTILE = 3,
+} splatt_tile_type;
+
+typedef enum splatt_decomp_type {
+  SPLATT_DECOMP_COARSE = 0,
+  SPLATT_DECOMP_MEDIUM = 1,
+  SPLATT_DECOMP_FINE = 2,
+} splatt_decomp_type;
+
+typedef enum splatt_comm_type {
+  SPLATT_COMM_POINT2POINT = 0,
+  SPLATT_COMM_ALL2ALL = 1,
+} splatt_comm_type;
+
 typedef enum splatt_option_type {
-  SPLATT_OPTION_TOLERANCE = 0,  /* convergence tolerance (1e-5)       */
+  SPLATT_OPTION_NTHREADS = 0,   /* worker threads (0 = hw default)    */
+  SPLATT_OPTION_TOLERANCE,      /* convergence tolerance (1e-5)       */
+  SPLATT_OPTION_REGULARIZE,     /* ridge term on the Gram diagonal (0)*/
   SPLATT_OPTION_NITER,          /* max ALS iterations (50)            */
   SPLATT_OPTION_VERBOSITY,      /* splatt_verbosity_type (LOW)        */
-  SPLATT_OPTION_NTHREADS,       /* OpenMP threads (omp default)       */
   SPLATT_OPTION_RANDSEED,       /* RNG seed (fixed default)           */
   SPLATT_OPTION_CSF_ALLOC,      /* splatt_csf_type (TWOMODE)          */
-  SPLATT_OPTION_REGULARIZE,     /* ridge term on the Gram diagonal (0)  */
-  SPLATT_OPTION_TILE,           /* advisory: gather-range bucketing     */
+  SPLATT_OPTION_TILE,           /* splatt_tile_type (NOTILE)            */
   SPLATT_OPTION_TILELEVEL,      /* advisory (GPU build picks LDS tiles) */
   SPLATT_OPTION_PRIVTHRESH,     /* advisory (GPU folds runs in regs)    */
-  SPLATT_OPTION_DECOMP,         /* advisory (Python layer selects)      */
-  SPLATT_OPTION_COMM,           /* advisory (RCCL collectives)          */
+  SPLATT_OPTION_DECOMP,         /* splatt_decomp_type (MEDIUM)          */
+  SPLATT_OPTION_COMM,           /* splatt_comm_type (ALL2ALL analog)    */
   SPLATT_OPTION_NOPTIONS
 } splatt_option_type;
 

@@ -154,12 +154,77 @@ void bin_write(const SpTensor<V> & tt, const std::string & path,
   }
 }
 
+// Reader for the reference's .bin layout (reference src/io.h:71-88 bin_header,
+// io.c:161-195 p_tt_read_binary_file): int32 magic (0 = BIN_COORD), then
+// uint64 idx_width and uint64 val_width (4 or 8), then nmodes, dims[nmodes]
+// and nnz at idx_width, then per-mode index arrays and the value array.
+// Indices are stored 0-based (the reference writes its internal labels).
+template <typename V>
+SpTensor<V> bin_read_ref(FILE * fp, const std::string & path) {
+  uint64_t iw = 0, vw = 0;
+  if (fread(&iw, 8, 1, fp) != 1 || fread(&vw, 8, 1, fp) != 1 ||
+      (iw != 4 && iw != 8) || (vw != 4 && vw != 8))
+    throw std::runtime_error("bad reference binary header in " + path);
+  auto rd_idx = [&](uint64_t * out, uint64_t n) {
+    if (iw == 8) {
+      if (fread(out, 8, n, fp) != n)
+        throw std::runtime_error("truncated " + path);
+    } else {
+      std::vector<uint32_t> v32(n);
+      if (fread(v32.data(), 4, n, fp) != n)
+        throw std::runtime_error("truncated " + path);
+      for (uint64_t i = 0; i < n; ++i) out[i] = v32[i];
+    }
+  };
+  uint64_t nm = 0, nnz = 0, dims64[MAX_NMODES];
+  rd_idx(&nm, 1);
+  if (nm < 1 || nm > MAX_NMODES)
+    throw std::runtime_error("bad nmodes in " + path);
+  rd_idx(dims64, nm);
+  rd_idx(&nnz, 1);
+  idx_t dims[MAX_NMODES];
+  for (uint64_t m = 0; m < nm; ++m) dims[m] = (idx_t)dims64[m];
+  SpTensor<V> tt((int)nm, nnz, dims);
+  std::vector<uint64_t> col(nnz);
+  for (uint64_t m = 0; m < nm; ++m) {
+    rd_idx(col.data(), nnz);
+    for (uint64_t i = 0; i < nnz; ++i) tt.ind[m][i] = (idx_t)col[i];
+  }
+  if (vw == 4) {
+    std::vector<float> vf(nnz);
+    if (fread(vf.data(), 4, nnz, fp) != nnz)
+      throw std::runtime_error("truncated values in " + path);
+    for (idx_t i = 0; i < nnz; ++i) tt.vals[i] = (V)vf[i];
+  } else {
+    std::vector<double> vd(nnz);
+    if (fread(vd.data(), 8, nnz, fp) != nnz)
+      throw std::runtime_error("truncated values in " + path);
+    for (idx_t i = 0; i < nnz; ++i) tt.vals[i] = (V)vd[i];
+  }
+  return tt;
+}
+
 template <typename V>
 SpTensor<V> bin_read(const std::string & path) {
   FilePtr f = xopen(path, "rb");
   char magic[8];
-  if (fread(magic, 1, 8, f.get()) != 8 || memcmp(magic, BIN_MAGIC, 8) != 0)
+  if (fread(magic, 1, 8, f.get()) != 8)
     throw std::runtime_error("bad binary tensor magic in " + path);
+  if (memcmp(magic, BIN_MAGIC, 8) != 0) {
+    // not ours: try the reference layout (int32 magic 0=COORD/1=CSF,
+    // then two uint64 widths); rewind past only the int32
+    int32_t rmagic;
+    memcpy(&rmagic, magic, 4);
+    if (rmagic == 0) {
+      if (fseek(f.get(), 4, SEEK_SET) != 0)
+        throw std::runtime_error("seek failed in " + path);
+      return bin_read_ref<V>(f.get(), path);
+    }
+    if (rmagic == 1)
+      throw std::runtime_error("reference BIN_CSF files are not supported; "
+                               "convert from coordinate form: " + path);
+    throw std::runtime_error("bad binary tensor magic in " + path);
+  }
   uint32_t ib = 0, vb = 0;
   uint64_t nm = 0, nnz = 0;
   if (fread(&ib, 4, 1, f.get()) != 1 || fread(&vb, 4, 1, f.get()) != 1 ||

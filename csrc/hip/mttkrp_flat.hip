@@ -407,12 +407,18 @@ mttkrp_flat4_kern(const int32_t * __restrict__ key,
 // lane = column (chunked by 64), wave walks its span serially. Correctness
 // path for ranks outside the spec set.
 // up to 7 other modes (8-mode tensors, reference SPLATT_MAX_NMODES);
-// idx/mats arrive as a small device-side pointer table
+// idx/mats arrive BY VALUE in the kernel-arg struct (no device-side table,
+// no upload: the launch stays async and hipGraph-capturable)
+template <typename V>
+struct PtrTab {
+  const int32_t * idx[8];
+  const V * mats[8];
+};
+
 template <typename V, int NOTHER>
 __global__ void __launch_bounds__(256)
 mttkrp_flat_generic_kern(const int32_t * __restrict__ key,
-                         const int32_t * const * __restrict__ idx,
-                         const V * const * __restrict__ mats,
+                         const PtrTab<V> tab,
                          const V * __restrict__ vals, int64_t nnz,
                          int64_t span, int rank, V * __restrict__ out) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -424,7 +430,7 @@ mttkrp_flat_generic_kern(const int32_t * __restrict__ key,
   const int32_t * ip[NOTHER];
   const V * mp[NOTHER];
   #pragma unroll
-  for (int t = 0; t < NOTHER; ++t) { ip[t] = idx[t]; mp[t] = mats[t]; }
+  for (int t = 0; t < NOTHER; ++t) { ip[t] = tab.idx[t]; mp[t] = tab.mats[t]; }
   for (int cb = 0; cb < rank; cb += WAVE) {
     const int c = cb + lane;
     if (c >= rank) break;
@@ -511,21 +517,11 @@ void launch_flat(const int32_t * key, const int32_t * const idx[8],
 #undef LU
 #undef L1
 #undef ARGS
-  // generic path (odd ranks or >5 modes): pointer tables shipped by value
-  // through constant kernel args would need 2x8 slots; simplest is a tiny
-  // device-side table uploaded per call on the same stream.
-  static thread_local void * d_tab = nullptr;
-  if (!d_tab) (void)hipMalloc(&d_tab, 16 * sizeof(void*));
-  void * h_tab[16];
-  for (int t = 0; t < 8; ++t) h_tab[t] = (void*)(t < nother ? idx[t] : nullptr);
-  for (int t = 0; t < 8; ++t) h_tab[8 + t] = (void*)(t < nother ? mats[t] : nullptr);
-  // the prior user of d_tab may still be queued on `st`: order the update
-  // behind it, and use a blocking copy (pageable h_tab is a stack array)
-  (void)hipStreamSynchronize(st);
-  (void)hipMemcpy(d_tab, h_tab, sizeof(h_tab), hipMemcpyHostToDevice);
-  const int32_t * const * dixp = (const int32_t * const *)d_tab;
-  const V * const * dmp = (const V * const *)((void**)d_tab + 8);
-#define GARGS key, dixp, dmp, vals, nnz, span, rank, out
+  // generic path (odd ranks or >5 modes): pointer tables travel by value
+  // in the kernel-arg block — async launch, graph-capturable, no cleanup
+  PtrTab<V> tab{};
+  for (int t = 0; t < nother; ++t) { tab.idx[t] = idx[t]; tab.mats[t] = mats[t]; }
+#define GARGS key, tab, vals, nnz, span, rank, out
   switch (nother) {
     case 2:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 2>), grid, block, 0, st, GARGS); break;
     case 3:  hipLaunchKernelGGL((mttkrp_flat_generic_kern<V, 3>), grid, block, 0, st, GARGS); break;

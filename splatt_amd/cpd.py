@@ -98,6 +98,15 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
         if os.path.exists(opts.checkpoint_path):
             ck = torch.load(opts.checkpoint_path, map_location=dev,
                             weights_only=True)
+            meta = ck.get("meta")
+            want = {"dims": list(dims), "rank": rank, "dtype": str(dtype),
+                    "seed": opts.seed}
+            if meta is not None and meta != want:
+                bad = {k: (meta.get(k), want[k]) for k in want
+                       if meta.get(k) != want[k]}
+                raise ValueError(
+                    f"checkpoint {opts.checkpoint_path} does not match this "
+                    f"run (checkpoint vs current): {bad}")
             factors = [f.to(dev) for f in ck["factors"]]
             grams = [gram(f) for f in factors]
             lam = ck["lambda"].to(dev)
@@ -149,7 +158,10 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             tmp = opts.checkpoint_path + ".tmp"
             torch.save({"factors": [f.cpu() for f in factors],
                         "lambda": lam.cpu(), "iteration": it, "fit": fit,
-                        "fit_trace": trace}, tmp)
+                        "fit_trace": trace,
+                        "meta": {"dims": list(dims), "rank": rank,
+                                 "dtype": str(dtype), "seed": opts.seed}},
+                       tmp)
             os.replace(tmp, opts.checkpoint_path)
         if it > 0 and abs(fit - old_fit) < opts.tolerance:
             break

@@ -29,10 +29,19 @@ def test_version(lib):
     assert lib.splatt_version_minor() == 1
 
 
+# reference numeric slot/return values (types_config.h:103-215):
+OPT_NTHREADS, OPT_TOL, OPT_REG, OPT_NITER, OPT_VERB = 0, 1, 2, 3, 4
+SUCCESS = 1
+
+
 def test_default_opts(lib):
     o = lib.splatt_default_opts()
-    assert o[0] == pytest.approx(1e-5)   # tolerance
-    assert o[1] == 50                    # niter
+    assert o[OPT_TOL] == pytest.approx(1e-5)
+    assert o[OPT_NITER] == 50
+    assert o[OPT_NTHREADS] == 0
+    assert o[OPT_VERB] == 1     # SPLATT_VERBOSITY_LOW
+    assert o[6] == 1            # SPLATT_OPTION_CSF_ALLOC = TWOMODE
+    assert o[10] == 1           # SPLATT_OPTION_DECOMP = MEDIUM
     lib.splatt_free_opts(o)
 
 
@@ -43,7 +52,7 @@ def test_csf_load_and_cpd(lib, tmp_path, small3):
     nmodes = ctypes.c_uint64()
     csf = ctypes.c_void_p()
     rc = lib.splatt_csf_load(tns, ctypes.byref(nmodes), ctypes.byref(csf), o)
-    assert rc == 0
+    assert rc == SUCCESS
     assert nmodes.value == 3
     assert lib.splatt_csf_nnz(csf) == small3.nnz
 
@@ -56,9 +65,9 @@ def test_csf_load_and_cpd(lib, tmp_path, small3):
                     ("fit", ctypes.c_double)]
 
     k = Kruskal()
-    o[1] = 5  # niter
+    o[OPT_NITER] = 5
     rc = lib.splatt_cpd_als(csf, 8, o, ctypes.byref(k))
-    assert rc == 0
+    assert rc == SUCCESS
     assert k.nmodes == 3
     assert 0.0 <= k.fit < 1.0
     # fit must match the Python CPU driver exactly (same seed/init)
@@ -76,7 +85,8 @@ def test_capi_mttkrp(lib, tmp_path, small3):
     o = lib.splatt_default_opts()
     nmodes = ctypes.c_uint64()
     csf = ctypes.c_void_p()
-    assert lib.splatt_csf_load(tns, ctypes.byref(nmodes), ctypes.byref(csf), o) == 0
+    assert lib.splatt_csf_load(tns, ctypes.byref(nmodes),
+                               ctypes.byref(csf), o) == SUCCESS
     rank = 8
     mats = [sp.seeded_init(d, rank, m, 1).contiguous()
             for m, d in enumerate(small3.dims)]
@@ -87,7 +97,7 @@ def test_capi_mttkrp(lib, tmp_path, small3):
     rc = lib.splatt_mttkrp(1, rank, csf, ptrs,
                            ctypes.cast(out.data_ptr(),
                                        ctypes.POINTER(ctypes.c_double)), o)
-    assert rc == 0
+    assert rc == SUCCESS
     ref = sp.mttkrp_stream(small3, mats, 1)
     assert (out - ref).abs().max() < 1e-10
     lib.splatt_free_csf(csf, o)
@@ -125,7 +135,7 @@ def test_csf_convert_in_memory(lib, small3):
                                 vals.ctypes.data_as(
                                     ctypes.POINTER(ctypes.c_double)),
                                 ctypes.byref(csf), o)
-    assert rc == 0
+    assert rc == SUCCESS
     assert lib.splatt_csf_nnz(csf) == nnz
     lib.splatt_free_csf(csf, o)
     lib.splatt_free_opts(o)

@@ -112,3 +112,19 @@ def test_cpd_regularize():
     # host reference path honors it too
     kn = sp.cpd_als_cpu_native(t, 6, ob)
     assert kn.fit < sp.cpd_als_cpu_native(t, 6, o0).fit + 1e-12
+
+
+def test_checkpoint_mismatch_rejected(tmp_path, small3):
+    """Resuming from a checkpoint written by a different rank/seed run
+    raises a clear error instead of silently continuing."""
+    import pytest as _pytest
+    ck = str(tmp_path / "ck.pt")
+    sp.cpd_als(small3, 8, sp.CpdOptions(max_iters=2, tolerance=0.0,
+                                        checkpoint_path=ck))
+    with _pytest.raises(ValueError, match="does not match"):
+        sp.cpd_als(small3, 6, sp.CpdOptions(max_iters=4, tolerance=0.0,
+                                            checkpoint_path=ck, resume=True))
+    with _pytest.raises(ValueError, match="seed"):
+        sp.cpd_als(small3, 8, sp.CpdOptions(max_iters=4, tolerance=0.0,
+                                            seed=123, checkpoint_path=ck,
+                                            resume=True))

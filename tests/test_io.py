@@ -114,3 +114,40 @@ def test_tns_accepts_comments_and_blanks(tmp_path):
     f.write_text("# header comment\n\n1 2 3 4.0\n% other comment\n2 3 1 5.0\n")
     t = sp.load(str(f))
     assert t.nnz == 2 and t.dims == [2, 3, 3]
+
+
+@pytest.mark.parametrize("iw,vw", [(4, 4), (4, 8), (8, 4), (8, 8)])
+def test_reference_bin_format_read(tmp_path, small3, iw, vw):
+    """Files in the REFERENCE repo's .bin layout (int32 magic=BIN_COORD,
+    u64 idx/val widths, then nmodes/dims/nnz + arrays at those widths;
+    reference src/io.h:71-88, io.c:161-195) load transparently."""
+    import struct
+    p = tmp_path / "ref.bin"
+    it = {4: "I", 8: "Q"}[iw]
+    vt = {4: "f", 8: "d"}[vw]
+    nnz = small3.nnz
+    with open(p, "wb") as f:
+        f.write(struct.pack("<i", 0))             # SPLATT_BIN_COORD
+        f.write(struct.pack("<QQ", iw, vw))
+        f.write(struct.pack(f"<{it}", 3))         # nmodes
+        f.write(struct.pack(f"<3{it}", *small3.dims))
+        f.write(struct.pack(f"<{it}", nnz))
+        for m in range(3):
+            f.write(struct.pack(f"<{nnz}{it}",
+                                *small3.inds[m].tolist()))
+        f.write(struct.pack(f"<{nnz}{vt}", *small3.vals.tolist()))
+    t2 = sp.load(p)
+    assert t2.dims == small3.dims
+    assert torch.equal(t2.inds, small3.inds)
+    tol = 1e-6 if vw == 4 else 0.0
+    assert (t2.vals - small3.vals).abs().max() <= tol
+
+
+def test_reference_bin_csf_rejected(tmp_path):
+    import struct
+    p = tmp_path / "csf.bin"
+    with open(p, "wb") as f:
+        f.write(struct.pack("<i", 1))             # SPLATT_BIN_CSF
+        f.write(struct.pack("<QQ", 8, 8))
+    with pytest.raises(RuntimeError, match="BIN_CSF"):
+        sp.load(p)
