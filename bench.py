@@ -13,8 +13,12 @@ Launch (driver contract):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
 
-Weak scaling: the partitioned mode's dimension grows with N and every rank
-generates (and owns) one full-size layer shard, so per-GPU work is fixed.
+Default experiment = BASELINE.json config 4: Amazon-Reviews-shaped
+(4.8M x 1.8M x 1.8M, 1.74B nnz) rank-16 f64 CPD-ALS, medium-grained
+nmodes-D grid over the N GPUs at FIXED global nnz — STRONG scaling, so the
+driver's N=1 BENCH and the N=1,2,4,8 SCALE sweep measure one experiment
+(config 5 = `--config delicious4d`). `--decomp coarse` gives the weak-
+scaling variant (per-GPU work fixed, partitioned mode grows with N).
 """
 from __future__ import annotations
 
@@ -69,10 +73,18 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--config", default="nell2", choices=list(CONFIGS))
+    # default = BASELINE.json config 4: Amazon-shaped rank-16 CPD-ALS,
+    # medium-grained over N GPUs at FIXED global nnz (strong scaling), so
+    # the driver's N=1 BENCH and N=1..8 SCALE runs are one experiment.
+    # SPLATT_BENCH_* env overrides let CI exercise the default argv at
+    # CPU-sized shapes (tests/test_bench_dist.py).
+    ap.add_argument("--config",
+                    default=os.environ.get("SPLATT_BENCH_CONFIG", "amazon"),
+                    choices=list(CONFIGS))
     ap.add_argument("--rank-f", type=int, default=0, help="override CP rank")
     ap.add_argument("--dtype", default="f64", choices=["f64", "f32"])
-    ap.add_argument("--device", default="cuda")
+    ap.add_argument("--device",
+                    default="cuda" if torch.cuda.is_available() else "cpu")
     ap.add_argument("--csf", default="all", choices=["one", "two", "all"])
     ap.add_argument("--synth", default="uniform", choices=["uniform", "zipf"],
                     help="index distribution of the synthetic tensor")
@@ -85,9 +97,13 @@ def main():
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise-reproducible kernels (SPLATT_DETERMINISTIC=1;"
                          " ~60%% of default throughput)")
-    ap.add_argument("--decomp", default="coarse", choices=["coarse", "medium"],
-                    help="coarse = 1D layers on the longest mode (weak "
-                         "scaling); medium = nmodes-D grid (strong scaling)")
+    ap.add_argument("--decomp",
+                    default=os.environ.get("SPLATT_BENCH_DECOMP", "medium"),
+                    choices=["coarse", "medium"],
+                    help="medium = nmodes-D grid, fixed global tensor "
+                         "(strong scaling; the BASELINE configs 4/5 shape); "
+                         "coarse = 1D layers on the longest mode, per-rank "
+                         "work fixed (weak scaling)")
     args = ap.parse_args()
 
     if args.deterministic:

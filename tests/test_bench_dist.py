@@ -50,3 +50,28 @@ def test_cli_cpd_torchrun_world2(tmp_path):
     assert "Final fit:" in r.stdout
     for m in range(3):
         assert (tmp_path / f"mode{m + 1}.mat").exists()
+
+
+@pytest.mark.timeout(600)
+def test_bench_driver_argv_world8(tmp_path):
+    """The driver's EXACT SCALE argv at N=8 (no extra flags) over gloo with
+    the RCCL reduce-scatter/all-gather primitives forced on — end-to-end
+    coverage of the default medium-grain strong-scaling path the 8-GPU
+    run executes. Only the tensor size is overridden (env),
+    never the code path."""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               SPLATT_BENCH_CONFIG="small",
+               SPLATT_FORCE_RS_PRIMS="1")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--nnodes=1", "--nproc-per-node", "8", "--local-addr", "127.0.0.1",
+         "bench.py", "--gpus", "8", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, cwd=ROOT, env=env, timeout=580)
+    assert r.returncode == 0, r.stderr[-1500:]
+    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    j = json.loads(lines[0])
+    assert j["n_gpus"] == 8
+    assert j["scaling"] == "strong"
+    assert "medium-grid" in j["config"]["parallelism"]
+    assert 0.0 <= j["config"]["fit"] < 1.0
