@@ -165,7 +165,7 @@ def main():
                          seed=0x5EED)
     from splatt_amd.parallel.grid import comm_stats
     cstats = comm_stats(dec, nnz_local, rank_f,
-                        8 if args.dtype == "f64" else 4)
+                        8 if args.dtype == "f64" else 4, cs=cs)
     if rank == 0:
         print(f"# comm: {cstats}", file=sys.stderr, flush=True)
     st = grid_cpd_init(cs, dec, rank_f, opts)

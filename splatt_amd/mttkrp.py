@@ -48,12 +48,20 @@ def _gpu_mttkrp_csf(c: Csf, depth: int, mats: List[torch.Tensor],
         c.vals, ma.contiguous(), mb.contiguous(), out, stream)
 
 
-def _stage_blocks(c: Csf) -> dict:
+def _stage_blocks(c: Csf, rows: tuple | None = None) -> dict:
     """Per-workgroup (nnz range, bucket row0) descriptors for the
-    LDS-staged kernel; cached on the Csf."""
-    cached = getattr(c, "_stage_blocks", None)
-    if cached is not None:
-        return cached
+    LDS-staged kernel; cached on the Csf. `rows=(lo,hi)` restricts the
+    descriptors to stream positions whose OUTPUT key (root label) lies in
+    [lo, hi) — the stream is bucket-major but root-key-sorted inside each
+    bucket, so the restriction is one searchsorted per bucket (done once,
+    cached; enables the chunked comm/compute pipeline in parallel/grid.py)."""
+    cache = getattr(c, "_stage_block_cache", None)
+    if cache is None:
+        cache = {}
+        object.__setattr__(c, "_stage_block_cache", cache)
+    ck = rows
+    if ck in cache:
+        return cache[ck]
     st = c._stage  # type: ignore[attr-defined]
     lvl, chunk, tiles = st["level"], st["chunk"], st["nbuckets"]
     dev = c.device
@@ -66,10 +74,25 @@ def _stage_blocks(c: Csf) -> dict:
             right=False).cpu()
     else:
         bnd = torch.tensor([0, nnz])
+    segs = []          # (start, end, bucket_id) stream segments to cover
+    if rows is None:
+        for b in range(len(bnd) - 1):
+            segs.append((int(bnd[b]), int(bnd[b + 1]), b))
+    else:
+        lo, hi = rows
+        key = c.ancestor_expand(0)
+        kb = torch.tensor([lo, hi], dtype=key.dtype, device=dev)
+        for b in range(len(bnd) - 1):
+            s, e = int(bnd[b]), int(bnd[b + 1])
+            if s == e:
+                continue
+            pos = torch.searchsorted(key[s:e], kb).cpu()
+            p0, p1 = s + int(pos[0]), s + int(pos[1])
+            if p0 < p1:
+                segs.append((p0, p1, b))
     tgt = max(4096, nnz // int(os.environ.get("SPLATT_LDS_BLOCKS", "16384")))
     starts, ends, row0s = [], [], []
-    for b in range(len(bnd) - 1):
-        s, e = int(bnd[b]), int(bnd[b + 1])
+    for s, e, b in segs:
         p = s
         while p < e:
             q = min(e, p + tgt)
@@ -84,8 +107,38 @@ def _stage_blocks(c: Csf) -> dict:
         "chunk": chunk,
         "level": lvl,
     }
-    object.__setattr__(c, "_stage_blocks", blocks)
+    cache[ck] = blocks
     return blocks
+
+
+def _key_sorted(c: Csf, depth: int) -> bool:
+    """Is the per-nnz output-key expansion non-decreasing? (Cached; true
+    for root-output streams of unbucketed builds.)"""
+    cache = getattr(c, "_key_sorted_cache", None)
+    if cache is None:
+        cache = {}
+        object.__setattr__(c, "_key_sorted_cache", cache)
+    if depth not in cache:
+        key = c.ancestor_expand(depth)
+        cache[depth] = bool((key[1:] >= key[:-1]).all()) if key.numel() > 1 \
+            else True
+    return cache[depth]
+
+
+def _key_bounds(c: Csf, depth: int, lo: int, hi: int) -> tuple:
+    """Stream positions [p0,p1) whose sorted output key is in [lo,hi)
+    (cached ints — the one-time device sync happens on first use)."""
+    cache = getattr(c, "_key_bounds_cache", None)
+    if cache is None:
+        cache = {}
+        object.__setattr__(c, "_key_bounds_cache", cache)
+    k = (depth, lo, hi)
+    if k not in cache:
+        key = c.ancestor_expand(depth)
+        kb = torch.tensor([lo, hi], dtype=key.dtype, device=key.device)
+        pos = torch.searchsorted(key, kb).cpu()
+        cache[k] = (int(pos[0]), int(pos[1]))
+    return cache[k]
 
 
 def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
@@ -136,22 +189,26 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
     native().gpu_mttkrp_flat_det(key, idx, ms, vals, out, ws, stream)
 
 
+def _use_lds(c: Csf, depth: int, rank: int) -> bool:
+    return (depth == 0 and c.nmodes <= 5
+            and getattr(c, "_stage", None) is not None
+            and rank in (4, 8, 16, 32, 64)
+            and os.environ.get("SPLATT_NO_LDS") != "1")
+
+
 def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
-                     out: torch.Tensor) -> None:
+                     out: torch.Tensor, rows: tuple | None = None) -> None:
     """Flat expanded-CSF kernel (see csrc/hip/mttkrp_flat.hip): per-nnz
     product of the non-output modes' rows folded by runs of the output
     key. Root-output dispatch on bucketed builds uses the LDS-staged
-    kernel (csrc/hip/mttkrp_lds.hip) when the rank is in the spec set."""
+    kernel (csrc/hip/mttkrp_lds.hip) when the rank is in the spec set.
+    `rows=(lo,hi)` restricts the launch to output rows in [lo,hi)."""
     nm = c.nmodes
     key = c.ancestor_expand(depth)
     rank = int(mats[0].shape[1])
     stream = torch.cuda.current_stream().cuda_stream
-    use_lds = (depth == 0 and nm <= 5
-               and getattr(c, "_stage", None) is not None
-               and rank in (4, 8, 16, 32, 64)
-               and os.environ.get("SPLATT_NO_LDS") != "1")
-    if use_lds:
-        blocks = _stage_blocks(c)
+    if _use_lds(c, depth, rank):
+        blocks = _stage_blocks(c, rows)
         lvl = blocks["level"]
         idx = [c.ancestor_expand(lvl)]
         ms = [mats[c.dim_perm[lvl]].contiguous()]
@@ -160,24 +217,53 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
                 continue
             idx.append(c.ancestor_expand(l))
             ms.append(mats[c.dim_perm[l]].contiguous())
+        if int(blocks["start"].numel()) == 0:
+            return
         native().gpu_mttkrp_flat5(
             key, idx, ms, c.vals, blocks["start"], blocks["end"],
             blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[lvl]],
             out, stream)
         return
+    p0, p1 = 0, c.nnz
+    if rows is not None:
+        if not _key_sorted(c, depth):
+            raise ValueError("rows-restricted MTTKRP needs a key-sorted "
+                             "stream for this mode")
+        p0, p1 = _key_bounds(c, depth, rows[0], rows[1])
+        if p0 >= p1:
+            return
     idx, ms = [], []
     for l in range(nm):
         if l == depth:
             continue
-        idx.append(c.ancestor_expand(l))
+        idx.append(c.ancestor_expand(l)[p0:p1])
         ms.append(mats[c.dim_perm[l]].contiguous())
-    native().gpu_mttkrp_flat(key, idx, ms, c.vals, out, stream)
+    native().gpu_mttkrp_flat(key[p0:p1], idx, ms, c.vals[p0:p1], out, stream)
+
+
+def mttkrp_rows_ok(src: CsfSet | Csf, mode: int, rank: int) -> bool:
+    """Can mttkrp() honor a rows=(lo,hi) restriction for this mode?
+    True when the dispatch lands on the LDS-staged kernel or on a
+    key-sorted stream (any root-output ALLMODE build). The chunked
+    comm/compute pipeline (parallel/grid.py) probes this and falls back
+    to the unchunked schedule otherwise."""
+    if isinstance(src, CsfSet):
+        c = src.csfs[src.mode_csf[mode]]
+        depth = src.mode_depth[mode]
+    else:
+        c, depth = src, src.level_of_mode(mode)
+    if os.environ.get("SPLATT_DETERMINISTIC") == "1":
+        return False
+    if c.device.type == "cuda" and _use_lds(c, depth, rank):
+        return True
+    return _key_sorted(c, depth)
 
 
 def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
            out: Optional[torch.Tensor] = None,
            nthreads: int = 0, alg: str = "flat",
-           deterministic: Optional[bool] = None) -> torch.Tensor:
+           deterministic: Optional[bool] = None,
+           rows: Optional[tuple] = None) -> torch.Tensor:
     """MTTKRP for output `mode`; `mats` indexed by tensor mode.
 
     Device algorithms: 'flat' (default, expanded-CSF streaming kernel) or
@@ -185,6 +271,9 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
     keeps multiple MTTKRP algorithms selectable the same way (bench.c).
     `deterministic` (or SPLATT_DETERMINISTIC=1) selects the
     bitwise-reproducible device kernel (no atomics; ALLMODE + spec ranks).
+    `rows=(lo,hi)`: compute ONLY output rows in [lo,hi) (zeroing just that
+    slice of `out`) — the building block of the chunked reduce-scatter
+    pipeline; probe support with mttkrp_rows_ok().
     """
     if isinstance(src, CsfSet):
         c = src.csfs[src.mode_csf[mode]]
@@ -213,15 +302,43 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
     if c.device.type == "cuda":
         if native().hip_arch() != 950:
             raise RuntimeError("HIP kernels not built for gfx950")
-        out.zero_()
         if deterministic is None:
             deterministic = os.environ.get("SPLATT_DETERMINISTIC") == "1"
+        if rows is not None:
+            if deterministic or alg != "flat":
+                raise ValueError("rows-restricted MTTKRP supports the "
+                                 "default flat kernels only")
+            out[rows[0]: rows[1]].zero_()
+            _gpu_mttkrp_flat(c, depth, mats, out, rows)
+            return out
+        out.zero_()
         if deterministic:
             _gpu_mttkrp_det(c, depth, mats, out)
         elif alg == "flat":
             _gpu_mttkrp_flat(c, depth, mats, out)
         else:
             _gpu_mttkrp_csf(c, depth, mats, mode, out)
+        return out
+    if rows is not None:
+        # host rows-restricted path (multi-process gloo tests): same math
+        # via the sorted stream expansions + index_add_
+        lo, hi = rows
+        if not _key_sorted(c, depth):
+            raise ValueError("rows-restricted MTTKRP needs a key-sorted "
+                             "stream for this mode")
+        p0, p1 = _key_bounds(c, depth, lo, hi)
+        out[lo:hi].zero_()
+        if p0 < p1:
+            key = c.ancestor_expand(depth)[p0:p1].long()
+            x = None
+            for l in range(c.nmodes):
+                if l == depth:
+                    continue
+                il = c.ancestor_expand(l)[p0:p1].long()
+                r = mats[c.dim_perm[l]].index_select(0, il)
+                x = r.clone() if x is None else x.mul_(r)
+            x.mul_(c.vals[p0:p1].unsqueeze(1))
+            out.index_add_(0, key, x)
         return out
     res = native().mttkrp_csf_cpu(c.to_dict(), [m.cpu() for m in mats],
                                   mode, nthreads)

@@ -26,7 +26,7 @@ import torch.distributed as dist
 
 from splatt_amd.cpd import CpdOptions, Kruskal, seeded_init
 from splatt_amd.csf import CsfSet
-from splatt_amd.mttkrp import mttkrp
+from splatt_amd.mttkrp import mttkrp, mttkrp_rows_ok
 from splatt_amd.ops.dense import gram, solve_rows, spd_inverse
 from splatt_amd.sptensor import SpTensor
 
@@ -230,9 +230,9 @@ def _reduce_scatter_rows(full: torch.Tensor, lo: int, hi: int,
     return out[: hi - lo], work
 
 
-def _all_gather_rows(own: torch.Tensor, n: int, group, gsize: int,
-                     out: torch.Tensor) -> None:
-    """Concatenate the group's owned row blocks back into `out` (n x F)."""
+def _ag_rows_start(own: torch.Tensor, n: int, group, gsize: int):
+    """Begin an async all-gather of the group's owned row blocks; returns
+    a handle for _ag_rows_finish."""
     per = (n + gsize - 1) // gsize
     F = own.shape[1]
     src = own
@@ -241,13 +241,39 @@ def _all_gather_rows(own: torch.Tensor, n: int, group, gsize: int,
         src[: own.shape[0]] = own
     if _backend_is_nccl():
         buf = torch.empty(per * gsize, F, dtype=own.dtype, device=own.device)
-        dist.all_gather_into_tensor(buf, src.contiguous(), group=group)
-    else:
-        parts = [torch.empty(per, F, dtype=own.dtype, device=own.device)
-                 for _ in range(gsize)]
-        dist.all_gather(parts, src.contiguous(), group=group)
+        work = dist.all_gather_into_tensor(buf, src.contiguous(),
+                                           group=group, async_op=True)
+        return (work, buf, None)
+    parts = [torch.empty(per, F, dtype=own.dtype, device=own.device)
+             for _ in range(gsize)]
+    work = dist.all_gather(parts, src.contiguous(), group=group,
+                           async_op=True)
+    return (work, None, parts)
+
+
+def _ag_rows_finish(handle, n: int, out: torch.Tensor) -> None:
+    work, buf, parts = handle
+    if work is not None:
+        work.wait()
+    if buf is None:
         buf = torch.cat(parts, 0)
     out.copy_(buf[:n])
+
+
+def _all_gather_rows(own: torch.Tensor, n: int, group, gsize: int,
+                     out: torch.Tensor) -> None:
+    """Concatenate the group's owned row blocks back into `out` (n x F)."""
+    _ag_rows_finish(_ag_rows_start(own, n, group, gsize), n, out)
+
+
+def _comm_chunks(default: int = 4) -> int:
+    """Pipeline depth for the chunked MTTKRP -> reduce-scatter overlap
+    (SPLATT_COMM_CHUNKS; 1 disables)."""
+    import os
+    try:
+        return max(1, int(os.environ.get("SPLATT_COMM_CHUNKS", default)))
+    except ValueError:
+        return default
 
 
 @dataclass
@@ -313,32 +339,56 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
     use_rsag = _os.environ.get("SPLATT_NO_RSAG") != "1"
     for m in range(nm):
         mb = st.buf[: dec.chunkn[m]]
-        with tm("MTTKRP"):
-            mttkrp(st.cs, st.factors, m, out=mb)
         group = dec.layer_groups.get(m)
         distributed = _world() > 1 and group != "solo" and dec.repl(m) > 1
+        nrows = dec.chunkn[m]
+        # chunked pipeline depth: C partial MTTKRPs by output-row range,
+        # each chunk's collective issued as soon as its rows are done and
+        # overlapped with the next chunk's MTTKRP (the side-stream overlap
+        # the reference lacks — its loop is fully synchronous,
+        # mpi/mpi_cpd.c:704-748). RCCL runs the collective on its own HIP
+        # stream ordered after the producing kernel.
+        C = 1
+        if distributed and overlap:
+            C = _comm_chunks()
+            if C > 1 and (nrows < C * dec.repl(m)
+                          or not mttkrp_rows_ok(st.cs, m, F)):
+                C = 1
+        bounds = [i * nrows // C for i in range(C + 1)]
         if distributed and use_rsag:
             # the SURVEY §2.4 mapping: reduce-scatter partial rows to
-            # contiguous owners, solve/normalize/gram ONLY owned rows,
-            # all-gather the updated blocks (replaces alltoallv pair).
-            # The Gram Hadamard + inverse run UNDER the async collective.
+            # contiguous owners (per chunk), solve/normalize/gram ONLY
+            # owned rows, all-gather the updated blocks (replaces the
+            # reference's alltoallv pair). Gram Hadamard + SPD inverse
+            # also run UNDER the async collectives.
             gsize = dec.repl(m)
             my = dec.layer_ranks[m].index(dist.get_rank())
-            nrows = dec.chunkn[m]
-            per = (nrows + gsize - 1) // gsize
-            lo = min(my * per, nrows)
-            hi = min(lo + per, nrows)
-            with tm("COMM-RS"):
-                own_mb, work = _reduce_scatter_rows(mb, lo, hi, group, gsize)
+            owns, rs_works = [], []
+            for i in range(C):
+                clo, chi = bounds[i], bounds[i + 1]
+                with tm("MTTKRP"):
+                    mttkrp(st.cs, st.factors, m, out=mb,
+                           rows=(clo, chi) if C > 1 else None)
+                ni = chi - clo
+                per = (ni + gsize - 1) // gsize
+                olo = min(my * per, ni)
+                ohi = min(olo + per, ni)
+                with tm("COMM-RS"):
+                    own_i, w = _reduce_scatter_rows(mb[clo:chi], olo, ohi,
+                                                    group, gsize)
+                owns.append(own_i)
+                rs_works.append(w)
             with tm("SOLVE"):
                 G = torch.ones(F, F, dtype=dtype, device=dev)
                 for o in range(nm):
                     if o != m:
                         G *= st.grams[o]
                 Ginv = spd_inverse(G)
-            if work is not None:
-                with tm("COMM-WAIT"):
-                    work.wait()
+            with tm("COMM-WAIT"):
+                for w in rs_works:
+                    if w is not None:
+                        w.wait()
+            own_mb = owns[0] if C == 1 else torch.cat(owns, 0)
             A_own = solve_rows(own_mb, Ginv)
             # lambda over GLOBAL rows (owned rows are globally unique)
             if it == 0:
@@ -351,34 +401,57 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
                 lam = lam.clamp_(min=1.0)
             lam = torch.where(lam == 0, torch.ones_like(lam), lam)
             A_own = A_own / lam
+            A = torch.empty(nrows, F, dtype=dtype, device=dev)
+            # all-gathers start first; gram/fit partials + their small
+            # all-reduces run underneath, then the gathers are drained
+            ag = []
+            with tm("COMM-AG"):
+                ofs = 0
+                for i in range(C):
+                    ni = bounds[i + 1] - bounds[i]
+                    nown = min(ni, (my + 1) * ((ni + gsize - 1) // gsize))                         - min(ni, my * ((ni + gsize - 1) // gsize))
+                    ag.append(_ag_rows_start(A_own[ofs: ofs + nown], ni,
+                                             group, gsize))
+                    ofs += nown
             g = gram(A_own) if A_own.numel() else                 torch.zeros(F, F, dtype=dtype, device=dev)
             _ar(g)
-            A = torch.empty(nrows, F, dtype=dtype, device=dev)
-            with tm("COMM-AG"):
-                _all_gather_rows(A_own, nrows, group, gsize, A)
-            st.lam = lam
-            st.factors[m] = A
-            st.grams[m] = g
             if m == nm - 1:
                 # fit inner from OWNED rows (each global row owned exactly
                 # once across the job): partial now, summed at fit time
                 part = (own_mb.double() * A_own.double()).sum(dim=0)                     if A_own.numel() else torch.zeros(F, dtype=torch.float64,
                                                       device=dev)
                 st._rs_inner = part  # type: ignore[attr-defined]
+            with tm("COMM-AG"):
+                for i in range(C):
+                    _ag_rows_finish(ag[i], bounds[i + 1] - bounds[i],
+                                    A[bounds[i]: bounds[i + 1]])
+            st.lam = lam
+            st.factors[m] = A
+            st.grams[m] = g
             continue
-        work = None
+        works = []
         if distributed:
-            with tm("COMM-POST"):
-                work = dist.all_reduce(mb, group=group, async_op=True)
+            for i in range(C):
+                clo, chi = bounds[i], bounds[i + 1]
+                with tm("MTTKRP"):
+                    mttkrp(st.cs, st.factors, m, out=mb,
+                           rows=(clo, chi) if C > 1 else None)
+                with tm("COMM-POST"):
+                    works.append(dist.all_reduce(mb[clo:chi], group=group,
+                                                 async_op=True))
+        else:
+            with tm("MTTKRP"):
+                mttkrp(st.cs, st.factors, m, out=mb)
         with tm("SOLVE"):
             G = torch.ones(F, F, dtype=dtype, device=dev)
             for o in range(nm):
                 if o != m:
                     G *= st.grams[o]
             Ginv = spd_inverse(G)
-        if work is not None:
-            with tm("COMM-WAIT"):
-                work.wait()
+        with tm("COMM-WAIT"):
+            for w in works:
+                if w is not None:
+                    w.wait()
         with tm("SOLVE"):
             A = solve_rows(mb, Ginv)
         # lambda over GLOBAL rows of mode m
@@ -430,13 +503,18 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
 
 
 def comm_stats(dec: GridDecomp, shard_nnz: int, rank_f: int,
-               val_bytes: int = 8) -> dict:
+               val_bytes: int = 8, cs: Optional[CsfSet] = None) -> dict:
     """Per-rank communication volume per ALS iteration + nnz balance
     (reference mpi_rank_stats / mpi_cpd_stats, stats.c:298-465).
-    COLLECTIVE: every rank must call this (it all-reduces nnz counts)."""
+    COLLECTIVE: every rank must call this (it all-reduces nnz counts).
+    With `cs`, also reports rows actually TOUCHED by local nonzeros vs
+    rows EXCHANGED (the whole contiguous chunk): the contiguous-chunk
+    RS/AG trades need-based lists (reference ineed, mpi_setup.c:13-155)
+    for true reduce-scatter — this keeps that trade measured."""
     world = _world()
     per_mode = []
     total = 0
+    touched = []
     for m in range(len(dec.global_dims)):
         if world > 1 and dec.repl(m) > 1:
             # ring all-reduce over the layer: ~2x payload per member
@@ -446,6 +524,14 @@ def comm_stats(dec: GridDecomp, shard_nnz: int, rank_f: int,
             vol = 0
         per_mode.append(vol)
         total += vol
+        if cs is not None:
+            c = cs.csfs[cs.mode_csf[m]]
+            lab = c.ancestor_expand(cs.mode_depth[m])
+            tr = torch.tensor([float(torch.unique(lab).numel())],
+                              dtype=torch.float64, device=lab.device)
+            if world > 1:
+                dist.all_reduce(tr, op=dist.ReduceOp.MAX)
+            touched.append(int(tr.item()))
     nnz_t = torch.tensor([float(shard_nnz)], dtype=torch.float64)
     if world > 1:
         mx = nnz_t.clone()
@@ -455,9 +541,15 @@ def comm_stats(dec: GridDecomp, shard_nnz: int, rank_f: int,
         imbalance = float(mx) / (float(sm) / world) - 1.0
     else:
         imbalance = 0.0
-    return {"grid": dec.grid, "comm_bytes_per_iter": total,
-            "comm_bytes_per_mode": per_mode,
-            "nnz_imbalance": round(imbalance, 4)}
+    out = {"grid": dec.grid, "comm_bytes_per_iter": total,
+           "comm_bytes_per_mode": per_mode,
+           "nnz_imbalance": round(imbalance, 4)}
+    if cs is not None:
+        out["touched_rows_per_mode"] = touched
+        out["exchanged_rows_per_mode"] = [
+            dec.chunkn[m] if per_mode[m] else 0
+            for m in range(len(dec.global_dims))]
+    return out
 
 
 def grid_cpd_als(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
