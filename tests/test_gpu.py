@@ -445,3 +445,49 @@ def test_gpu_cpd_regularize():
     k0 = sp.cpd_als(sp.csf_alloc(t.to("cuda"), "all"), 8, o0)
     kb = sp.cpd_als(sp.csf_alloc(t.to("cuda"), "all"), 8, ob)
     assert kb.fit < k0.fit and kb.fit == kb.fit
+
+
+@pytest.mark.parametrize("staged", [False, True])
+@pytest.mark.parametrize("rank", [16, 32])
+def test_gpu_mttkrp_rows_restricted(t3, staged, rank):
+    """rows=(lo,hi) partial launches (the chunked comm pipeline's building
+    block) tile the full result exactly, on both the plain flat kernel
+    (key-sorted stream) and the LDS-staged bucketed build."""
+    from splatt_amd.mttkrp import mttkrp_rows_ok
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    cs = build_shard_csf(t3.to("cuda"), list(t3.dims), "all",
+                         flat_only=True, stage_rank=rank if staged else 0)
+    mats_c = make_mats(t3.dims, rank)
+    mats_g = [m.cuda() for m in mats_c]
+    for mode in range(3):
+        assert mttkrp_rows_ok(cs, mode, rank)
+        full = sp.mttkrp(cs, mats_g, mode)
+        out = torch.empty_like(full)
+        n = t3.dims[mode]
+        bounds = [0, n // 4, n // 2, (3 * n) // 4, n]
+        for i in range(4):
+            sp.mttkrp(cs, mats_g, mode, out=out,
+                      rows=(bounds[i], bounds[i + 1]))
+        err = (out - full).abs().max().item()
+        assert err < 1e-10, (staged, rank, mode, err)
+        ref = sp.mttkrp_stream(t3, mats_c, mode)
+        err2 = (out.cpu() - ref).abs().max().item()
+        assert err2 < 1e-8, (staged, rank, mode, err2)
+
+
+def test_gpu_generic_rank_graph_capturable(t3):
+    """The generic-rank fallback must be hipGraph-capturable (ADVICE r1:
+    no hipStreamSynchronize/blocking copies inside the launch path)."""
+    rank = 10   # not in the spec set -> generic kernel
+    cs = sp.csf_alloc(t3.to("cuda"), "all")
+    mats_g = make_mats(t3.dims, rank, device="cuda")
+    ref = sp.mttkrp(cs, mats_g, 0)
+    out = torch.empty_like(ref)
+    sp.mttkrp(cs, mats_g, 0, out=out)   # warm
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        sp.mttkrp(cs, mats_g, 0, out=out)
+    out.zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    assert (out - ref).abs().max().item() < 1e-10
