@@ -143,6 +143,128 @@ mttkrp_flat5_kern(const int32_t * __restrict__ key,
   atomic_add_g(&out[(int64_t)cur * F + c], acc);
 }
 
+// ------------------------------------------------- packed-stream v6
+// v5 loads key/i0/i1(/i2) as 3-4 separate 4-B stream reads per nonzero
+// slot; the per-CU TA address path is the measured binder
+// (profiles/ROUND1_PROFILING.md), so v6 packs them into ONE int4 word
+// per nonzero (built at CSF-build time, splatt_amd/csf.py): word layout
+// x = output key (root label), y = staged-level label, z/w = remaining
+// levels. Stream tag lookups per slot drop from 4-5 to 2 (pack + vals).
+template <typename V, int F, int NOTHER>
+__global__ void __launch_bounds__(WPB * WAVE)
+mttkrp_flat6_kern(const int * __restrict__ pack_raw,
+                  const V * __restrict__ m0, const V * __restrict__ m1,
+                  const V * __restrict__ m2,
+                  const V * __restrict__ vals,
+                  const int64_t * __restrict__ blk_start,
+                  const int64_t * __restrict__ blk_end,
+                  const int32_t * __restrict__ blk_row0,
+                  int32_t chunk, int32_t dim0,
+                  V * __restrict__ out) {
+  using Pack = __attribute__((ext_vector_type(4))) int;
+  const Pack * __restrict__ pack = reinterpret_cast<const Pack *>(pack_raw);
+  constexpr int GB = 8;
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  V * smem = reinterpret_cast<V *>(smem_raw);
+
+  const int b = blockIdx.x;
+  const int64_t b0 = blk_start[b];
+  const int64_t b1 = blk_end[b];
+  const int32_t row0 = blk_row0[b];
+  const int nrows = (int)min64((int64_t)chunk, (int64_t)dim0 - row0);
+
+  {
+    constexpr int VEC = 16 / sizeof(V);
+    using Vec = __attribute__((ext_vector_type(VEC))) V;
+    const int nel = nrows * F;
+    const int nvec = nel / VEC;
+    const int tid = threadIdx.x;
+    const Vec * src = reinterpret_cast<const Vec *>(m0 + (int64_t)row0 * F);
+    Vec * dst = reinterpret_cast<Vec *>(smem);
+    for (int ve = tid; ve < nvec; ve += WPB * WAVE) dst[ve] = src[ve];
+    for (int t = nvec * VEC + tid; t < nel; t += WPB * WAVE)
+      smem[t] = m0[(int64_t)row0 * F + t];
+  }
+  __syncthreads();
+
+  constexpr int R = WAVE / F;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wv = threadIdx.x / WAVE;
+  const int c = lane % F;
+  const int g = lane / F;
+  const int gbase = g * F;
+  const int nsub = WPB * R;
+  const int sub = wv * R + g;
+  const int64_t total = b1 - b0;
+  const int64_t gsz = (total + nsub - 1) / nsub;
+  const int64_t p0 = min64(b1, b0 + sub * gsz);
+  const int64_t p1 = min64(b1, p0 + gsz);
+  if (p0 >= p1) return;
+
+  int32_t cur = pack[p0].x;
+  V acc = (V)0;
+  for (int64_t pb = p0; pb < p1; pb += F) {
+    const int nb = (int)min64((int64_t)F, p1 - pb);
+    const int64_t ps = pb + (c < nb ? c : nb - 1);
+    const Pack preg = __builtin_nontemporal_load(&pack[ps]);
+    const V vreg = ldnt(&vals[ps]);
+    for (int ub = 0; ub < nb; ub += GB) {
+      const int ne = nb - ub < GB ? nb - ub : GB;
+      int32_t kk[GB];
+      V vv[GB], a0[GB], a1[GB], a2[GB];
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        const int src = gbase + (u < ne ? ub + u : ub);
+        kk[u] = __shfl(preg.x, src, WAVE);
+        vv[u] = __shfl(vreg, src, WAVE);
+        const int32_t j0 = __shfl(preg.y, src, WAVE);
+        const int32_t j1 = __shfl(preg.z, src, WAVE);
+        a0[u] = smem[(j0 - row0) * F + c];          // LDS, no L1/TA
+        a1[u] = m1[(int64_t)j1 * F + c];
+        if (NOTHER > 2) {
+          const int32_t j2 = __shfl(preg.w, src, WAVE);
+          a2[u] = m2[(int64_t)j2 * F + c];
+        }
+      }
+      #pragma unroll
+      for (int u = 0; u < GB; ++u) {
+        if (u >= ne) break;
+        V x = vv[u] * a0[u] * a1[u];
+        if (NOTHER > 2) x *= a2[u];
+        if (kk[u] != cur) {
+          atomic_add_g(&out[(int64_t)cur * F + c], acc);
+          acc = (V)0;
+          cur = kk[u];
+        }
+        acc += x;
+      }
+    }
+  }
+  atomic_add_g(&out[(int64_t)cur * F + c], acc);
+}
+
+template <typename V>
+void launch_flat6(const int32_t * pack, const V * const mats[3],
+                  const V * vals, const int64_t * blk_start,
+                  const int64_t * blk_end, const int32_t * blk_row0,
+                  int64_t nblocks, int32_t chunk, int32_t dim0, V * out,
+                  int rank, int nother, hipStream_t st) {
+  dim3 grid((uint32_t)nblocks), block(WPB * WAVE);
+  const size_t lds = (size_t)chunk * rank * sizeof(V);
+#define ARGS6 pack, mats[0], mats[1], mats[2], vals, blk_start, blk_end, \
+              blk_row0, chunk, dim0, out
+#define L6(F_, N_) \
+  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_>), grid, block, lds, st, ARGS6)
+#define L6F(N_) \
+  switch (rank) { case 4: L6(4, N_); break; case 8: L6(8, N_); break; \
+                  case 16: L6(16, N_); break; case 32: L6(32, N_); break; \
+                  default: L6(64, N_); break; }
+  if (nother == 2) { L6F(2); } else { L6F(3); }
+#undef L6F
+#undef L6
+#undef ARGS6
+}
+
 template <typename V>
 void launch_flat5(const int32_t * key, const int32_t * const idx[4],
                   const V * const mats[4], const V * vals,
@@ -171,6 +293,30 @@ void launch_flat5(const int32_t * key, const int32_t * const idx[4],
 }
 
 }  // namespace
+
+extern "C" void splatt_hip_mttkrp_flat6_f64(
+    const int32_t * pack, const double * m0, const double * m1,
+    const double * m2, const double * vals, const int64_t * blk_start,
+    const int64_t * blk_end, const int32_t * blk_row0, int64_t nblocks,
+    int32_t chunk, int32_t dim0, double * out, int rank, int nother,
+    void * stream) {
+  const double * mats[3] = {m0, m1, m2};
+  launch_flat6<double>(pack, mats, vals, blk_start, blk_end, blk_row0,
+                       nblocks, chunk, dim0, out, rank, nother,
+                       (hipStream_t)stream);
+}
+
+extern "C" void splatt_hip_mttkrp_flat6_f32(
+    const int32_t * pack, const float * m0, const float * m1,
+    const float * m2, const float * vals, const int64_t * blk_start,
+    const int64_t * blk_end, const int32_t * blk_row0, int64_t nblocks,
+    int32_t chunk, int32_t dim0, float * out, int rank, int nother,
+    void * stream) {
+  const float * mats[3] = {m0, m1, m2};
+  launch_flat6<float>(pack, mats, vals, blk_start, blk_end, blk_row0,
+                      nblocks, chunk, dim0, out, rank, nother,
+                      (hipStream_t)stream);
+}
 
 extern "C" void splatt_hip_mttkrp_flat5_f64(
     const int32_t * key, const int32_t * i0, const int32_t * i1,

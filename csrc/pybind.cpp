@@ -425,6 +425,14 @@ void splatt_hip_mttkrp_flat5_f32(
     const int32_t*, const float*, const float*, const float*,
     const float*, const float*, const int64_t*, const int64_t*,
     const int32_t*, int64_t, int32_t, int32_t, float*, int, int, void*);
+void splatt_hip_mttkrp_flat6_f64(
+    const int32_t*, const double*, const double*, const double*,
+    const double*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    int32_t, int32_t, double*, int, int, void*);
+void splatt_hip_mttkrp_flat6_f32(
+    const int32_t*, const float*, const float*, const float*,
+    const float*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    int32_t, int32_t, float*, int, int, void*);
 }
 
 // LDS-staged variant: idx[0]/mats[0] is the bucketed level; block
@@ -458,6 +466,45 @@ static void py_gpu_mttkrp_flat5(Tensor key, std::vector<Tensor> idx,
     splatt_hip_mttkrp_flat5_f32(key.data_ptr<int32_t>(), ip[0], ip[1], ip[2],
                                 ip[3], mp[0], mp[1], mp[2], mp[3],
                                 vals.data_ptr<float>(),
+                                blk_start.data_ptr<int64_t>(),
+                                blk_end.data_ptr<int64_t>(),
+                                blk_row0.data_ptr<int32_t>(), nblocks,
+                                (int32_t)chunk, (int32_t)dim0,
+                                out.data_ptr<float>(), rank, nother,
+                                (void*)stream);
+  }
+}
+
+// packed-stream LDS variant: one int4 word per nonzero
+// (x = output key, y = staged level, z/w = remaining levels)
+static void py_gpu_mttkrp_flat6(Tensor pack, std::vector<Tensor> mats,
+                                Tensor vals, Tensor blk_start,
+                                Tensor blk_end, Tensor blk_row0,
+                                int64_t chunk, int64_t dim0, Tensor out,
+                                int64_t stream) {
+  const int nother = (int)mats.size();
+  TORCH_CHECK(nother >= 2 && nother <= 3);
+  TORCH_CHECK(pack.is_contiguous() && pack.size(1) == 4
+              && pack.scalar_type() == torch::kInt32,
+              "pack must be contiguous [nnz,4] int32");
+  const int rank = (int)mats[0].size(1);
+  const int64_t nblocks = blk_start.numel();
+  if (vals.scalar_type() == torch::kFloat64) {
+    const double * mp[3] = {nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
+    splatt_hip_mttkrp_flat6_f64(pack.data_ptr<int32_t>(), mp[0], mp[1],
+                                mp[2], vals.data_ptr<double>(),
+                                blk_start.data_ptr<int64_t>(),
+                                blk_end.data_ptr<int64_t>(),
+                                blk_row0.data_ptr<int32_t>(), nblocks,
+                                (int32_t)chunk, (int32_t)dim0,
+                                out.data_ptr<double>(), rank, nother,
+                                (void*)stream);
+  } else {
+    const float * mp[3] = {nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+    splatt_hip_mttkrp_flat6_f32(pack.data_ptr<int32_t>(), mp[0], mp[1],
+                                mp[2], vals.data_ptr<float>(),
                                 blk_start.data_ptr<int64_t>(),
                                 blk_end.data_ptr<int64_t>(),
                                 blk_row0.data_ptr<int32_t>(), nblocks,
@@ -649,6 +696,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   }, "Ginv = G^-1 for SPD FxF (F<=64), one-workgroup Cholesky");
   m.def("gpu_mttkrp_flat5", &py_gpu_mttkrp_flat5,
         "LDS-staged flat MTTKRP (bucketed builds, root output)");
+  m.def("gpu_mttkrp_flat6", &py_gpu_mttkrp_flat6,
+        "LDS-staged flat MTTKRP over the packed int4 stream");
   m.def("partition_weighted", [](std::vector<int64_t> w, int nparts) {
     int64_t bn = 0;
     auto parts = sp::partition_weighted(w.data(), (int64_t)w.size(), nparts, &bn);

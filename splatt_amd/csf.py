@@ -65,10 +65,20 @@ class Csf:
         )
         # carry expansions + stage metadata (a frozen flat CSF has no fptr
         # tree to recompute them from)
-        cache = getattr(self, "_expand_cache", None)
-        if cache:
+        pack = getattr(self, "_pack", None)
+        if pack is not None:
+            p = pack.to(device)
+            levels = list(self._pack_levels)  # type: ignore[attr-defined]
+            object.__setattr__(c, "_pack", p)
+            object.__setattr__(c, "_pack_levels", levels)
             object.__setattr__(c, "_expand_cache",
-                               {l: t.to(device) for l, t in cache.items()})
+                               {l: p[:, col] for col, l in enumerate(levels)})
+            c.fids[self.nmodes - 1] = c._expand_cache[self.nmodes - 1]  # type: ignore[attr-defined]
+        else:
+            cache = getattr(self, "_expand_cache", None)
+            if cache:
+                object.__setattr__(c, "_expand_cache",
+                                   {l: t.to(device) for l, t in cache.items()})
         stage = getattr(self, "_stage", None)
         if stage is not None:
             object.__setattr__(c, "_stage", dict(stage))
@@ -263,8 +273,25 @@ def _build_csf_device(t: SpTensor, perm: List[int],
     if flat_only:
         c = Csf(dims=list(t.dims), dim_perm=list(perm),
                 fptr=[None] * nm, fids=[None] * nm, vals=svals)
-        c.fids[nm - 1] = sinds[nm - 1].to(torch.int32)
-        cache = {l: sinds[l].to(torch.int32) for l in range(nm)}
+        import os as _os3
+        if (stage_meta is not None and nm <= 4
+                and _os3.environ.get("SPLATT_PACK") != "0"):
+            # packed stream: ONE int4 word per nonzero [key, staged level,
+            # remaining levels] for the v6 LDS kernel (halves the stream
+            # tag lookups, csrc/hip/mttkrp_lds.hip). The per-level
+            # expansions become strided VIEWS into the pack — no second
+            # copy of the index streams.
+            lvl = stage_meta["level"]
+            levels = [0, lvl] + [l for l in range(1, nm) if l != lvl]
+            pack = torch.zeros(nnz, 4, dtype=torch.int32, device=dev)
+            for col, l in enumerate(levels):
+                pack[:, col] = sinds[l].to(torch.int32)
+            cache = {l: pack[:, col] for col, l in enumerate(levels)}
+            object.__setattr__(c, "_pack", pack)
+            object.__setattr__(c, "_pack_levels", levels)
+        else:
+            cache = {l: sinds[l].to(torch.int32) for l in range(nm)}
+        c.fids[nm - 1] = cache[nm - 1]
         object.__setattr__(c, "_expand_cache", cache)
         if stage_meta is not None:
             object.__setattr__(c, "_stage", stage_meta)

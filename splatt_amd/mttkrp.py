@@ -86,7 +86,7 @@ def _stage_blocks(c: Csf, rows: tuple | None = None) -> dict:
             s, e = int(bnd[b]), int(bnd[b + 1])
             if s == e:
                 continue
-            pos = torch.searchsorted(key[s:e], kb).cpu()
+            pos = torch.searchsorted(key[s:e].contiguous(), kb).cpu()
             p0, p1 = s + int(pos[0]), s + int(pos[1])
             if p0 < p1:
                 segs.append((p0, p1, b))
@@ -134,7 +134,7 @@ def _key_bounds(c: Csf, depth: int, lo: int, hi: int) -> tuple:
         object.__setattr__(c, "_key_bounds_cache", cache)
     k = (depth, lo, hi)
     if k not in cache:
-        key = c.ancestor_expand(depth)
+        key = c.ancestor_expand(depth).contiguous()
         kb = torch.tensor([lo, hi], dtype=key.dtype, device=key.device)
         pos = torch.searchsorted(key, kb).cpu()
         cache[k] = (int(pos[0]), int(pos[1]))
@@ -176,6 +176,11 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
             key = key.index_select(0, perm).contiguous()
             idx = [i.index_select(0, perm).contiguous() for i in idx]
             vals = vals.index_select(0, perm).contiguous()
+        else:
+            # packed builds expose strided views; the det kernel wants
+            # contiguous streams
+            key = key.contiguous()
+            idx = [i.contiguous() for i in idx]
         streams[depth] = (key, idx, vals)
     key, idx, vals = streams[depth]
     ms = [mats[c.dim_perm[l]].contiguous() for l in range(nm) if l != depth]
@@ -210,15 +215,26 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
     if _use_lds(c, depth, rank):
         blocks = _stage_blocks(c, rows)
         lvl = blocks["level"]
-        idx = [c.ancestor_expand(lvl)]
+        if int(blocks["start"].numel()) == 0:
+            return
+        pack = getattr(c, "_pack", None)
+        if pack is not None and os.environ.get("SPLATT_NO_PACK") != "1":
+            # packed-stream v6: word order [key, staged, rest] fixed at
+            # build time (csf.py); mats follow _pack_levels[1:]
+            levels = c._pack_levels  # type: ignore[attr-defined]
+            ms = [mats[c.dim_perm[l]].contiguous() for l in levels[1:]]
+            native().gpu_mttkrp_flat6(
+                pack, ms, c.vals, blocks["start"], blocks["end"],
+                blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[lvl]],
+                out, stream)
+            return
+        idx = [c.ancestor_expand(lvl).contiguous()]
         ms = [mats[c.dim_perm[lvl]].contiguous()]
         for l in range(nm):
             if l in (depth, lvl):
                 continue
-            idx.append(c.ancestor_expand(l))
+            idx.append(c.ancestor_expand(l).contiguous())
             ms.append(mats[c.dim_perm[l]].contiguous())
-        if int(blocks["start"].numel()) == 0:
-            return
         native().gpu_mttkrp_flat5(
             key, idx, ms, c.vals, blocks["start"], blocks["end"],
             blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[lvl]],
@@ -236,9 +252,10 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
     for l in range(nm):
         if l == depth:
             continue
-        idx.append(c.ancestor_expand(l)[p0:p1])
+        idx.append(c.ancestor_expand(l)[p0:p1].contiguous())
         ms.append(mats[c.dim_perm[l]].contiguous())
-    native().gpu_mttkrp_flat(key[p0:p1], idx, ms, c.vals[p0:p1], out, stream)
+    native().gpu_mttkrp_flat(key[p0:p1].contiguous(), idx, ms,
+                             c.vals[p0:p1], out, stream)
 
 
 def mttkrp_rows_ok(src: CsfSet | Csf, mode: int, rank: int) -> bool:
