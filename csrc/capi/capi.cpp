@@ -19,6 +19,13 @@
 
 using namespace splatt;
 
+namespace splatt {
+// device engine (csrc/capi/capi_gpu.cpp, linked with the HIP kernels)
+bool capi_gpu_available();
+template <typename V>
+Kruskal<V> cpd_als_gpu(const CsfSet<V> &, int, const Options &);
+}
+
 struct splatt_csf {
   CsfSet<double> set;
 };
@@ -118,7 +125,11 @@ int splatt_cpd_als(const splatt_csf * tensors, splatt_idx_t nfactors,
                    const double * options, splatt_kruskal * factored) {
   try {
     auto opt = opts_from_array(options);
-    auto k = cpd_als(tensors->set, (int)nfactors, opt);
+    // HIP engine when an MI355X is visible (SPLATT_CAPI_CPU=1 forces the
+    // host core); errors surface loudly rather than silently falling back
+    auto k = capi_gpu_available()
+                 ? cpd_als_gpu<double>(tensors->set, (int)nfactors, opt)
+                 : cpd_als(tensors->set, (int)nfactors, opt);
     factored->rank = nfactors;
     factored->nmodes = (splatt_idx_t)k.nmodes;
     factored->fit = k.fit;
@@ -169,6 +180,10 @@ splatt_mttkrp_ws * splatt_mttkrp_alloc_ws(const splatt_csf * tensors,
   return ws;
 }
 void splatt_mttkrp_free_ws(splatt_mttkrp_ws * ws) { aligned_free64(ws); }
+
+int splatt_gpu_available(void) {
+  return splatt::capi_gpu_available() ? 1 : 0;
+}
 
 int splatt_version_major(void) { return 0; }
 int splatt_version_minor(void) { return 1; }

@@ -147,6 +147,10 @@ splatt_mttkrp_ws * splatt_mttkrp_alloc_ws(const splatt_csf * tensors,
 void splatt_mttkrp_free_ws(splatt_mttkrp_ws * ws);
 
 /* ------------------------------------------------------------- version */
+/** 1 when an AMD GPU is visible and the HIP engine will serve
+ * splatt_cpd_als (override with SPLATT_CAPI_CPU=1). */
+int splatt_gpu_available(void);
+
 int splatt_version_major(void);
 int splatt_version_minor(void);
 int splatt_version_subminor(void);

@@ -354,6 +354,86 @@ extern "C" void splatt_hip_gram_det_f32(const float * A, int64_t n, int F,
                      (hipStream_t)stream, Gpart, nparts, F, G);
 }
 
+// ---- generic column reductions/scales for the torch-free C API device
+// driver (csrc/capi/capi_gpu.cpp): F <= 64, one lane group per column.
+namespace {
+
+__device__ inline void atomic_max_f64(double * p, double v) {
+  unsigned long long * u = reinterpret_cast<unsigned long long *>(p);
+  unsigned long long old = *u, assumed;
+  while (__longlong_as_double(old) < v) {
+    assumed = old;
+    old = atomicCAS(u, assumed, __double_as_longlong(v));
+    if (old == assumed) break;
+  }
+}
+
+// which: 0 = sum of squares, 1 = max(|x|); out[F] pre-initialized
+// (0 for sum, 0 for max — values are magnitudes)
+__global__ void __launch_bounds__(256)
+colacc_kern(const double * __restrict__ A, int64_t n, int F, int which,
+            double * __restrict__ out) {
+  const int gpc = (int)(blockDim.x / F);      // row groups per block
+  const int tid = (int)threadIdx.x;
+  if (tid >= gpc * F) return;
+  const int c = tid % F;
+  const int64_t g = (int64_t)blockIdx.x * gpc + tid / F;
+  const int64_t stride = (int64_t)gridDim.x * gpc;
+  double acc = 0.0;
+  for (int64_t i = g; i < n; i += stride) {
+    const double x = A[i * F + c];
+    if (which == 0) acc += x * x;
+    else acc = fmax(acc, fabs(x));
+  }
+  if (which == 0) atomic_add_g(&out[c], acc);
+  else atomic_max_f64(&out[c], acc);
+}
+
+__global__ void __launch_bounds__(256)
+colscale_kern(double * __restrict__ A, int64_t n, int F,
+              const double * __restrict__ lam) {
+  const int64_t t0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t e = t0; e < n * F; e += stride)
+    A[e] /= lam[e % F];
+}
+
+// out[f] += sum_i X[i,f] * A[i,f]  (fit inner product; out pre-zeroed)
+__global__ void __launch_bounds__(256)
+coldot_kern(const double * __restrict__ X, const double * __restrict__ A,
+            int64_t n, int F, double * __restrict__ out) {
+  const int gpc = (int)(blockDim.x / F);
+  const int tid = (int)threadIdx.x;
+  if (tid >= gpc * F) return;
+  const int c = tid % F;
+  const int64_t g = (int64_t)blockIdx.x * gpc + tid / F;
+  const int64_t stride = (int64_t)gridDim.x * gpc;
+  double acc = 0.0;
+  for (int64_t i = g; i < n; i += stride)
+    acc += X[i * F + c] * A[i * F + c];
+  atomic_add_g(&out[c], acc);
+}
+
+}  // namespace
+
+extern "C" void splatt_hip_colacc_f64(const double * A, int64_t n, int F,
+                                      int which, double * out,
+                                      void * stream) {
+  hipLaunchKernelGGL(colacc_kern, dim3(256), dim3(256), 0,
+                     (hipStream_t)stream, A, n, F, which, out);
+}
+extern "C" void splatt_hip_colscale_f64(double * A, int64_t n, int F,
+                                        const double * lam, void * stream) {
+  hipLaunchKernelGGL(colscale_kern, dim3(512), dim3(256), 0,
+                     (hipStream_t)stream, A, n, F, lam);
+}
+extern "C" void splatt_hip_coldot_f64(const double * X, const double * A,
+                                      int64_t n, int F, double * out,
+                                      void * stream) {
+  hipLaunchKernelGGL(coldot_kern, dim3(256), dim3(256), 0,
+                     (hipStream_t)stream, X, A, n, F, out);
+}
+
 extern "C" int splatt_hip_rowsolve_f64(const double * A, const double * B,
                                        double * C, int64_t n, int F,
                                        void * stream) {

@@ -50,28 +50,42 @@ def build_hip_objects():
     return objs
 
 
-def build_native_lib():
-    """libsplatt.so (C API) + the `splatt` host CLI binary — torch-free."""
+def build_native_lib(hip_objs):
+    """libsplatt.so (C API, torch-free) + the `splatt` host CLI binary.
+    The HIP kernel objects are linked in so splatt_cpd_als dispatches to
+    the device engine (csrc/capi/capi_gpu.cpp) when a GPU is visible."""
     bindir = ROOT / "bin"
     bindir.mkdir(exist_ok=True)
     lib = bindir / "libsplatt.so"
     exe = bindir / "splatt"
     srcs = [str(p) for p in CORE_SOURCES] + [str(ROOT / "csrc/capi/capi.cpp")]
+    gpu_src = ROOT / "csrc/capi/capi_gpu.cpp"
     newest = max(Path(s).stat().st_mtime
-                 for s in srcs + [str(ROOT / "csrc/capi/splatt.h"),
-                                  str(ROOT / "csrc/capi/splatt_main.cpp")])
+                 for s in srcs + hip_objs
+                 + [str(gpu_src), str(ROOT / "csrc/capi/splatt.h"),
+                    str(ROOT / "csrc/capi/splatt_main.cpp")])
     if not lib.exists() or lib.stat().st_mtime < newest:
+        # capi_gpu.cpp is host-only HIP-API code: g++ with the AMD platform
+        # macro (keeps one OpenMP runtime, libgomp, across libsplatt)
+        gpu_obj = ROOT / "build" / "hip_obj" / "capi_gpu.o"
+        subprocess.check_call(
+            ["g++", "-O3", "-std=c++17", "-fPIC", "-fopenmp",
+             "-D__HIP_PLATFORM_AMD__=1", f"-I{ROCM}/include",
+             f"-I{ROOT}/csrc", "-c", str(gpu_src), "-o", str(gpu_obj)])
         cxx = ["g++", "-O3", "-std=c++17", "-fPIC", "-fopenmp", "-march=native",
                f"-I{ROOT}/csrc"]
         print("[g++] libsplatt.so + splatt CLI", flush=True)
-        subprocess.check_call(cxx + ["-shared", "-o", str(lib)] + srcs)
+        subprocess.check_call(
+            cxx + ["-shared", "-o", str(lib)] + srcs + [str(gpu_obj)]
+            + hip_objs + [f"-L{ROCM}/lib", "-lamdhip64",
+                          f"-Wl,-rpath,{ROCM}/lib"])
         subprocess.check_call(
             cxx + ["-o", str(exe), str(ROOT / "csrc/capi/splatt_main.cpp"),
                    f"-L{bindir}", "-lsplatt", f"-Wl,-rpath,{bindir}"])
 
 
 hip_objs = build_hip_objects()
-build_native_lib()
+build_native_lib(hip_objs)
 
 ext = CppExtension(
     name="splatt_amd._C",

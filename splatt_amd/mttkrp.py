@@ -236,9 +236,9 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
             idx.append(c.ancestor_expand(l).contiguous())
             ms.append(mats[c.dim_perm[l]].contiguous())
         native().gpu_mttkrp_flat5(
-            key, idx, ms, c.vals, blocks["start"], blocks["end"],
-            blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[lvl]],
-            out, stream)
+            key.contiguous(), idx, ms, c.vals, blocks["start"],
+            blocks["end"], blocks["row0"], blocks["chunk"],
+            c.dims[c.dim_perm[lvl]], out, stream)
         return
     p0, p1 = 0, c.nnz
     if rows is not None:
