@@ -94,6 +94,12 @@ def main():
                     help="-1 (default): LDS-staged bucketing sized by "
                          "SPLATT_LDS_KB; 0: off; N>1: plain gather-range "
                          "buckets without LDS staging")
+    ap.add_argument("--factor-store", default="f64",
+                    choices=["f64", "f32", "bf16"],
+                    help="reduced-precision factor STORAGE for the MTTKRP "
+                         "gathers (accumulation stays f64). Documented "
+                         "experimental mode; the default and all headline "
+                         "numbers are full f64")
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise-reproducible kernels (SPLATT_DETERMINISTIC=1;"
                          " ~60%% of default throughput)")
@@ -108,6 +114,8 @@ def main():
 
     if args.deterministic:
         os.environ["SPLATT_DETERMINISTIC"] = "1"
+    if args.factor_store != "f64":
+        os.environ["SPLATT_FACTOR_STORE"] = args.factor_store
     dims, nnz_shard, rank_f, _ = CONFIGS[args.config]
     if args.rank_f:
         rank_f = args.rank_f
@@ -242,7 +250,8 @@ def main():
             "higher_is_better": True,
             "scaling": scaling,
             "vs_baseline": None,
-            "dtype": args.dtype,
+            "dtype": (args.dtype if args.factor_store == "f64" else
+                      f"{args.dtype}+{args.factor_store}-store"),
             "data": "synthetic",
             "config": {
                 "model": f"{args.config}-shaped CPD-ALS",

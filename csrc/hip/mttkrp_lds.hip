@@ -150,11 +150,35 @@ mttkrp_flat5_kern(const int32_t * __restrict__ key,
 // per nonzero (built at CSF-build time, splatt_amd/csf.py): word layout
 // x = output key (root label), y = staged-level label, z/w = remaining
 // levels. Stream tag lookups per slot drop from 4-5 to 2 (pack + vals).
-template <typename V, int F, int NOTHER>
+//
+// S = factor STORAGE type (defaults to the compute type V). S=float or
+// S=bf16 with V=double is the documented reduced-precision factor-store
+// mode for HBM-bound shapes: gathered rows shrink 2-4x in cache lines
+// while every multiply-accumulate stays f64 (ROADMAP item 2b).
+struct bf16 { uint16_t v; };
+template <typename S> struct VecElemT { using type = S; };
+template <> struct VecElemT<bf16> { using type = uint16_t; };
+__device__ __forceinline__ double to_compute(double x, double) { return x; }
+__device__ __forceinline__ float to_compute(float x, float) { return x; }
+__device__ __forceinline__ double to_compute(float x, double) {
+  return (double)x;
+}
+__device__ __forceinline__ double to_compute(bf16 x, double) {
+  union { uint32_t u; float f; } c;
+  c.u = (uint32_t)x.v << 16;
+  return (double)c.f;
+}
+__device__ __forceinline__ float to_compute(bf16 x, float) {
+  union { uint32_t u; float f; } c;
+  c.u = (uint32_t)x.v << 16;
+  return c.f;
+}
+
+template <typename V, int F, int NOTHER, typename S = V>
 __global__ void __launch_bounds__(WPB * WAVE)
 mttkrp_flat6_kern(const int * __restrict__ pack_raw,
-                  const V * __restrict__ m0, const V * __restrict__ m1,
-                  const V * __restrict__ m2,
+                  const S * __restrict__ m0, const S * __restrict__ m1,
+                  const S * __restrict__ m2,
                   const V * __restrict__ vals,
                   const int64_t * __restrict__ blk_start,
                   const int64_t * __restrict__ blk_end,
@@ -165,7 +189,7 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
   const Pack * __restrict__ pack = reinterpret_cast<const Pack *>(pack_raw);
   constexpr int GB = 8;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  V * smem = reinterpret_cast<V *>(smem_raw);
+  S * smem = reinterpret_cast<S *>(smem_raw);
 
   const int b = blockIdx.x;
   const int64_t b0 = blk_start[b];
@@ -174,16 +198,24 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
   const int nrows = (int)min64((int64_t)chunk, (int64_t)dim0 - row0);
 
   {
-    constexpr int VEC = 16 / sizeof(V);
-    using Vec = __attribute__((ext_vector_type(VEC))) V;
     const int nel = nrows * F;
-    const int nvec = nel / VEC;
     const int tid = threadIdx.x;
-    const Vec * src = reinterpret_cast<const Vec *>(m0 + (int64_t)row0 * F);
-    Vec * dst = reinterpret_cast<Vec *>(smem);
-    for (int ve = tid; ve < nvec; ve += WPB * WAVE) dst[ve] = src[ve];
-    for (int t = nvec * VEC + tid; t < nel; t += WPB * WAVE)
-      smem[t] = m0[(int64_t)row0 * F + t];
+    if ((((int64_t)row0 * F * sizeof(S)) & 15) == 0) {
+      // vectorize the stage copy through an integral proxy of the same
+      // width (struct element types cannot form ext_vector_type)
+      using E = typename VecElemT<S>::type;
+      constexpr int VEC = 16 / sizeof(E);
+      using Vec = __attribute__((ext_vector_type(VEC))) E;
+      const int nvec = nel / VEC;
+      const Vec * src = reinterpret_cast<const Vec *>(m0 + (int64_t)row0 * F);
+      Vec * dst = reinterpret_cast<Vec *>(smem);
+      for (int ve = tid; ve < nvec; ve += WPB * WAVE) dst[ve] = src[ve];
+      for (int t = nvec * VEC + tid; t < nel; t += WPB * WAVE)
+        smem[t] = m0[(int64_t)row0 * F + t];
+    } else {
+      for (int t = tid; t < nel; t += WPB * WAVE)
+        smem[t] = m0[(int64_t)row0 * F + t];
+    }
   }
   __syncthreads();
 
@@ -211,7 +243,8 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
     for (int ub = 0; ub < nb; ub += GB) {
       const int ne = nb - ub < GB ? nb - ub : GB;
       int32_t kk[GB];
-      V vv[GB], a0[GB], a1[GB], a2[GB];
+      V vv[GB];
+      S a0[GB], a1[GB], a2[GB];
       #pragma unroll
       for (int u = 0; u < GB; ++u) {
         const int src = gbase + (u < ne ? ub + u : ub);
@@ -229,8 +262,8 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
       #pragma unroll
       for (int u = 0; u < GB; ++u) {
         if (u >= ne) break;
-        V x = vv[u] * a0[u] * a1[u];
-        if (NOTHER > 2) x *= a2[u];
+        V x = vv[u] * to_compute(a0[u], (V)0) * to_compute(a1[u], (V)0);
+        if (NOTHER > 2) x *= to_compute(a2[u], (V)0);
         if (kk[u] != cur) {
           atomic_add_g(&out[(int64_t)cur * F + c], acc);
           acc = (V)0;
@@ -243,18 +276,18 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
   atomic_add_g(&out[(int64_t)cur * F + c], acc);
 }
 
-template <typename V>
-void launch_flat6(const int32_t * pack, const V * const mats[3],
+template <typename V, typename S = V>
+void launch_flat6(const int32_t * pack, const S * const mats[3],
                   const V * vals, const int64_t * blk_start,
                   const int64_t * blk_end, const int32_t * blk_row0,
                   int64_t nblocks, int32_t chunk, int32_t dim0, V * out,
                   int rank, int nother, hipStream_t st) {
   dim3 grid((uint32_t)nblocks), block(WPB * WAVE);
-  const size_t lds = (size_t)chunk * rank * sizeof(V);
+  const size_t lds = (size_t)chunk * rank * sizeof(S);
 #define ARGS6 pack, mats[0], mats[1], mats[2], vals, blk_start, blk_end, \
               blk_row0, chunk, dim0, out
 #define L6(F_, N_) \
-  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_>), grid, block, lds, st, ARGS6)
+  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S>), grid, block, lds, st, ARGS6)
 #define L6F(N_) \
   switch (rank) { case 4: L6(4, N_); break; case 8: L6(8, N_); break; \
                   case 16: L6(16, N_); break; case 32: L6(32, N_); break; \
@@ -316,6 +349,32 @@ extern "C" void splatt_hip_mttkrp_flat6_f32(
   launch_flat6<float>(pack, mats, vals, blk_start, blk_end, blk_row0,
                       nblocks, chunk, dim0, out, rank, nother,
                       (hipStream_t)stream);
+}
+
+// reduced-precision factor STORAGE (f64 accumulation) — ROADMAP 2b
+extern "C" void splatt_hip_mttkrp_flat6_f64f32(
+    const int32_t * pack, const float * m0, const float * m1,
+    const float * m2, const double * vals, const int64_t * blk_start,
+    const int64_t * blk_end, const int32_t * blk_row0, int64_t nblocks,
+    int32_t chunk, int32_t dim0, double * out, int rank, int nother,
+    void * stream) {
+  const float * mats[3] = {m0, m1, m2};
+  launch_flat6<double, float>(pack, mats, vals, blk_start, blk_end,
+                              blk_row0, nblocks, chunk, dim0, out, rank,
+                              nother, (hipStream_t)stream);
+}
+
+extern "C" void splatt_hip_mttkrp_flat6_f64bf16(
+    const int32_t * pack, const uint16_t * m0, const uint16_t * m1,
+    const uint16_t * m2, const double * vals, const int64_t * blk_start,
+    const int64_t * blk_end, const int32_t * blk_row0, int64_t nblocks,
+    int32_t chunk, int32_t dim0, double * out, int rank, int nother,
+    void * stream) {
+  const bf16 * mats[3] = {(const bf16*)m0, (const bf16*)m1,
+                          (const bf16*)m2};
+  launch_flat6<double, bf16>(pack, mats, vals, blk_start, blk_end,
+                             blk_row0, nblocks, chunk, dim0, out, rank,
+                             nother, (hipStream_t)stream);
 }
 
 extern "C" void splatt_hip_mttkrp_flat5_f64(

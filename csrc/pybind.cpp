@@ -433,6 +433,14 @@ void splatt_hip_mttkrp_flat6_f32(
     const int32_t*, const float*, const float*, const float*,
     const float*, const int64_t*, const int64_t*, const int32_t*, int64_t,
     int32_t, int32_t, float*, int, int, void*);
+void splatt_hip_mttkrp_flat6_f64f32(
+    const int32_t*, const float*, const float*, const float*,
+    const double*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    int32_t, int32_t, double*, int, int, void*);
+void splatt_hip_mttkrp_flat6_f64bf16(
+    const int32_t*, const uint16_t*, const uint16_t*, const uint16_t*,
+    const double*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    int32_t, int32_t, double*, int, int, void*);
 }
 
 // LDS-staged variant: idx[0]/mats[0] is the bucketed level; block
@@ -490,6 +498,35 @@ static void py_gpu_mttkrp_flat6(Tensor pack, std::vector<Tensor> mats,
   const int rank = (int)mats[0].size(1);
   const int64_t nblocks = blk_start.numel();
   if (vals.scalar_type() == torch::kFloat64) {
+    // reduced-precision factor STORAGE with f64 accumulation
+    if (mats[0].scalar_type() == torch::kFloat32) {
+      const float * mp[3] = {nullptr, nullptr, nullptr};
+      for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+      splatt_hip_mttkrp_flat6_f64f32(pack.data_ptr<int32_t>(), mp[0], mp[1],
+                                     mp[2], vals.data_ptr<double>(),
+                                     blk_start.data_ptr<int64_t>(),
+                                     blk_end.data_ptr<int64_t>(),
+                                     blk_row0.data_ptr<int32_t>(), nblocks,
+                                     (int32_t)chunk, (int32_t)dim0,
+                                     out.data_ptr<double>(), rank, nother,
+                                     (void*)stream);
+      return;
+    }
+    if (mats[0].scalar_type() == torch::kBFloat16) {
+      const uint16_t * mp[3] = {nullptr, nullptr, nullptr};
+      for (int t = 0; t < nother; ++t)
+        mp[t] = reinterpret_cast<const uint16_t*>(
+            mats[t].data_ptr<at::BFloat16>());
+      splatt_hip_mttkrp_flat6_f64bf16(pack.data_ptr<int32_t>(), mp[0],
+                                      mp[1], mp[2], vals.data_ptr<double>(),
+                                      blk_start.data_ptr<int64_t>(),
+                                      blk_end.data_ptr<int64_t>(),
+                                      blk_row0.data_ptr<int32_t>(), nblocks,
+                                      (int32_t)chunk, (int32_t)dim0,
+                                      out.data_ptr<double>(), rank, nother,
+                                      (void*)stream);
+      return;
+    }
     const double * mp[3] = {nullptr, nullptr, nullptr};
     for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
     splatt_hip_mttkrp_flat6_f64(pack.data_ptr<int32_t>(), mp[0], mp[1],

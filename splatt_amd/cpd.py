@@ -83,6 +83,10 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
     factors = [seeded_init(dims[m], rank, m, opts.seed, dtype=dtype).to(dev)
                for m in range(nm)]
     grams = [gram(f) for f in factors]
+    from splatt_amd.mttkrp import factor_store_dtype
+    qdt = factor_store_dtype() if (dev.type == "cuda"
+                                   and dtype == torch.float64) else None
+    qfactors = [f.to(qdt) for f in factors] if qdt else None
     norm_x = float(cs.csfs[0].vals.double().square().sum())
     lam = torch.ones(rank, dtype=dtype, device=dev)
     buf = torch.empty(max(dims), rank, dtype=dtype, device=dev)
@@ -109,6 +113,8 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
                     f"run (checkpoint vs current): {bad}")
             factors = [f.to(dev) for f in ck["factors"]]
             grams = [gram(f) for f in factors]
+            if qfactors is not None:
+                qfactors = [f.to(qdt) for f in factors]
             lam = ck["lambda"].to(dev)
             it0 = int(ck["iteration"]) + 1
             fit = old_fit = float(ck["fit"])
@@ -119,7 +125,8 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
         _t0 = _time.perf_counter()
         for m in range(nm):
             mb = buf[: dims[m]]
-            mttkrp(cs, factors, m, out=mb, nthreads=opts.nthreads)
+            mttkrp(cs, qfactors or factors, m, out=mb,
+                   nthreads=opts.nthreads)
             G = ones.clone()
             for o in range(nm):
                 if o != m:
@@ -133,6 +140,8 @@ def cpd_als(src: CsfSet | SpTensor, rank: int,
             A = solve_rows(mb, spd_inverse(G))
             lam = _normalize(A, it)
             factors[m] = A
+            if qfactors is not None:
+                qfactors[m] = A.to(qdt)
             grams[m] = gram(A)
 
         # fit from last mode's pre-solve MTTKRP output (reference trick):

@@ -237,7 +237,11 @@ def _build_csf_device(t: SpTensor, perm: List[int],
             # nnz runs, -28% at ~1 nnz runs).
             import os as _os
             min_run = int(_os.environ.get("SPLATT_STAGE_MIN_RUN", "8"))
-            vbytes = t.vals.element_size()
+            # LDS tile sized by the factor STORAGE width (f32/bf16 store
+            # modes fit 2-4x more rows per bucket)
+            vbytes = {"f32": 4, "bf16": 2}.get(
+                _os.environ.get("SPLATT_FACTOR_STORE", ""),
+                t.vals.element_size())
             chunk_cap = max(64, (lds_kb * 1024) // (stage_rank * vbytes))
             root_dim = max(1, t.dims[perm[0]])
             # note: even L2-resident factors profit from staging (the TA

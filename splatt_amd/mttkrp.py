@@ -258,6 +258,15 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
                              c.vals[p0:p1], out, stream)
 
 
+def factor_store_dtype():
+    """Optional reduced-precision factor STORAGE for the MTTKRP gathers
+    (SPLATT_FACTOR_STORE=f32|bf16; accumulation stays in the tensor's
+    dtype, f64). Cuts gathered cache lines 2-4x on HBM-bound shapes
+    (documented mode — the default and every headline number stay f64)."""
+    env = os.environ.get("SPLATT_FACTOR_STORE", "")
+    return {"f32": torch.float32, "bf16": torch.bfloat16}.get(env)
+
+
 def mttkrp_rows_ok(src: CsfSet | Csf, mode: int, rank: int) -> bool:
     """Can mttkrp() honor a rows=(lo,hi) restriction for this mode?
     True when the dispatch lands on the LDS-staged kernel or on a
@@ -304,17 +313,23 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
     if len(mats) != c.nmodes:
         raise ValueError(f"need {c.nmodes} factor matrices, got {len(mats)}")
     rank = int(mats[0].shape[1])
+    storage = (c.device.type == "cuda" and c.vals.dtype == torch.float64
+               and mats[0].dtype in (torch.float32, torch.bfloat16))
     for m, A in enumerate(mats):
         if A.shape != (c.dims[m], rank):
             raise ValueError(
                 f"factor {m} has shape {tuple(A.shape)}, expected "
                 f"({c.dims[m]}, {rank})")
-        if A.dtype != c.vals.dtype or A.device != c.device:
+        if (A.dtype != c.vals.dtype and not storage) or A.device != c.device:
             raise ValueError(
                 f"factor {m}: dtype/device {A.dtype}/{A.device} does not "
                 f"match tensor {c.vals.dtype}/{c.device}")
+    if storage and getattr(c, "_pack", None) is None:
+        raise ValueError(
+            "reduced-precision factor storage needs an LDS-staged packed "
+            "build (build with stage_rank=rank; see SPLATT_FACTOR_STORE)")
     if out is None:
-        out = torch.empty(c.dims[mode], rank, dtype=mats[0].dtype,
+        out = torch.empty(c.dims[mode], rank, dtype=c.vals.dtype,
                           device=mats[0].device)
     if c.device.type == "cuda":
         if native().hip_arch() != 950:

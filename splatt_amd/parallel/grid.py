@@ -285,6 +285,9 @@ class GridCpdState:
     lam: torch.Tensor
     buf: torch.Tensor
     norm_x: float
+    # reduced-precision factor-store copies for the MTTKRP gathers
+    # (SPLATT_FACTOR_STORE)
+    qfactors: Optional[List[torch.Tensor]] = None
     fit: float = 0.0
     old_fit: float = 0.0
     niters: int = 0
@@ -311,10 +314,14 @@ def grid_cpd_init(shard_cs: CsfSet, dec: GridDecomp, rank_f: int,
                       dtype=torch.float64, device=dev)
     _ar(nx)
     maxdim = max(dec.chunkn)
+    from splatt_amd.mttkrp import factor_store_dtype
+    qdt = factor_store_dtype() if (dev.type == "cuda"
+                                   and dtype == torch.float64) else None
     return GridCpdState(
         cs=shard_cs, dec=dec, factors=factors, grams=grams,
         lam=torch.ones(rank_f, dtype=dtype, device=dev),
         buf=torch.empty(maxdim, rank_f, dtype=dtype, device=dev),
+        qfactors=[f.to(qdt) for f in factors] if qdt else None,
         norm_x=float(nx.item()))
 
 
@@ -367,7 +374,7 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             for i in range(C):
                 clo, chi = bounds[i], bounds[i + 1]
                 with tm("MTTKRP"):
-                    mttkrp(st.cs, st.factors, m, out=mb,
+                    mttkrp(st.cs, st.qfactors or st.factors, m, out=mb,
                            rows=(clo, chi) if C > 1 else None)
                 ni = chi - clo
                 per = (ni + gsize - 1) // gsize
@@ -427,6 +434,8 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
                                     A[bounds[i]: bounds[i + 1]])
             st.lam = lam
             st.factors[m] = A
+            if st.qfactors is not None:
+                st.qfactors[m] = A.to(st.qfactors[m].dtype)
             st.grams[m] = g
             continue
         works = []
@@ -434,14 +443,14 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             for i in range(C):
                 clo, chi = bounds[i], bounds[i + 1]
                 with tm("MTTKRP"):
-                    mttkrp(st.cs, st.factors, m, out=mb,
+                    mttkrp(st.cs, st.qfactors or st.factors, m, out=mb,
                            rows=(clo, chi) if C > 1 else None)
                 with tm("COMM-POST"):
                     works.append(dist.all_reduce(mb[clo:chi], group=group,
                                                  async_op=True))
         else:
             with tm("MTTKRP"):
-                mttkrp(st.cs, st.factors, m, out=mb)
+                mttkrp(st.cs, st.qfactors or st.factors, m, out=mb)
         with tm("SOLVE"):
             G = torch.ones(F, F, dtype=dtype, device=dev)
             for o in range(nm):
@@ -470,6 +479,8 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
         A /= lam
         st.lam = lam
         st.factors[m] = A
+        if st.qfactors is not None:
+            st.qfactors[m] = A.to(st.qfactors[m].dtype)
         g = gram(A)
         if dec.grid[m] > 1:
             g /= dec.repl(m)

@@ -515,3 +515,46 @@ def test_gpu_packed_stream_matches(t3, rank):
             assert (out6 - out5).abs().max().item() < 1e-10
             ref = sp.mttkrp_stream(t, mats_c, mode)
             assert (out6.cpu() - ref).abs().max().item() < 1e-8
+
+
+@pytest.mark.parametrize("store", ["f32", "bf16"])
+def test_gpu_factor_store_mttkrp(t3, store):
+    """Reduced-precision factor STORAGE (SPLATT_FACTOR_STORE): gathers
+    read f32/bf16 rows, accumulation stays f64; result tracks the f64
+    oracle within storage precision (ROADMAP 2b documented mode)."""
+    import os
+    from splatt_amd.parallel.dist_cpd import build_shard_csf
+    rank = 16
+    os.environ["SPLATT_FACTOR_STORE"] = store
+    try:
+        cs = build_shard_csf(t3.to("cuda"), list(t3.dims), "all",
+                             flat_only=True, stage_rank=rank)
+        mats_c = make_mats(t3.dims, rank)
+        qdt = {"f32": torch.float32, "bf16": torch.bfloat16}[store]
+        mats_q = [m.cuda().to(qdt) for m in mats_c]
+        tol = {"f32": 5e-5, "bf16": 2e-1}[store]
+        for mode in range(3):
+            out = sp.mttkrp(cs, mats_q, mode)
+            assert out.dtype == torch.float64
+            ref = sp.mttkrp_stream(t3, mats_c, mode)
+            rel = ((out.cpu() - ref).abs().max()
+                   / ref.abs().max()).item()
+            assert rel < tol, (store, mode, rel)
+    finally:
+        del os.environ["SPLATT_FACTOR_STORE"]
+
+
+def test_gpu_factor_store_cpd_converges(t3):
+    """bf16-store CPD-ALS reaches a fit close to the f64 run."""
+    import os
+    ref = sp.cpd_als(sp.csf_alloc(t3.to("cuda"), "all"), 16,
+                     sp.CpdOptions(max_iters=4, tolerance=0.0))
+    os.environ["SPLATT_FACTOR_STORE"] = "bf16"
+    try:
+        from splatt_amd.parallel.dist_cpd import build_shard_csf
+        cs = build_shard_csf(t3.to("cuda"), list(t3.dims), "all",
+                             flat_only=True, stage_rank=16)
+        k = sp.cpd_als(cs, 16, sp.CpdOptions(max_iters=4, tolerance=0.0))
+    finally:
+        del os.environ["SPLATT_FACTOR_STORE"]
+    assert abs(k.fit - ref.fit) < 5e-3, (k.fit, ref.fit)
