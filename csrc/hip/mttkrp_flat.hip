@@ -23,7 +23,11 @@
 #include <cstdint>
 #include <cstdlib>
 
+#include "store_types.hpp"
+
 namespace {
+using splatt_store::bf16;
+using splatt_store::to_compute;
 
 constexpr int WAVE = 64;
 
@@ -126,15 +130,15 @@ mttkrp_flat_kern(const int32_t * __restrict__ key,
 // and redistributes them with ds_bpermute (__shfl); the factor-row gathers
 // then issue in independent batches of 8, giving ~4x the outstanding
 // gathers per wave at lower VGPR pressure.
-template <typename V, int F, int NOTHER, int GBP = 8>
+template <typename V, int F, int NOTHER, int GBP = 8, typename S = V>
 __global__ void __launch_bounds__(256)
 mttkrp_flat2_kern(const int32_t * __restrict__ key,
                   const int32_t * __restrict__ i0,
                   const int32_t * __restrict__ i1,
                   const int32_t * __restrict__ i2,
                   const int32_t * __restrict__ i3,
-                  const V * __restrict__ m0, const V * __restrict__ m1,
-                  const V * __restrict__ m2, const V * __restrict__ m3,
+                  const S * __restrict__ m0, const S * __restrict__ m1,
+                  const S * __restrict__ m2, const S * __restrict__ m3,
                   const V * __restrict__ vals, int64_t nnz, int64_t span,
                   V * __restrict__ out) {
   constexpr int R = WAVE / F;
@@ -167,7 +171,8 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
     for (int ub = 0; ub < nb; ub += GB) {
       const int ne = nb - ub < GB ? nb - ub : GB;         // group-uniform
       int32_t kk[GB];
-      V vv[GB], a0[GB], a1[GB], a2[GB], a3[GB];
+      V vv[GB];
+      S a0[GB], a1[GB], a2[GB], a3[GB];
       #pragma unroll
       for (int u = 0; u < GB; ++u) {
         const int src = gbase + (u < ne ? ub + u : ub);
@@ -189,9 +194,9 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
       #pragma unroll
       for (int u = 0; u < GB; ++u) {
         if (u >= ne) break;
-        V x = vv[u] * a0[u] * a1[u];
-        if (NOTHER > 2) x *= a2[u];
-        if (NOTHER > 3) x *= a3[u];
+        V x = vv[u] * to_compute(a0[u], (V)0) * to_compute(a1[u], (V)0);
+        if (NOTHER > 2) x *= to_compute(a2[u], (V)0);
+        if (NOTHER > 3) x *= to_compute(a3[u], (V)0);
         if (kk[u] != cur) {
           atomic_add_g(&out[(int64_t)cur * F + c], acc);
           acc = (V)0;
@@ -202,6 +207,37 @@ mttkrp_flat2_kern(const int32_t * __restrict__ key,
     }
   }
   atomic_add_g(&out[(int64_t)cur * F + c], acc);
+}
+
+// storage-typed launcher for the v2 kernel (the default spec path):
+// reduced-precision factor STORE for non-staged HBM-bound dispatches
+inline int64_t pick_span_decl(int64_t nnz);
+template <typename V, typename S>
+void launch_flat_store(const int32_t * key, const int32_t * const idx[8],
+                       const S * const mats[8], const V * vals, int64_t nnz,
+                       V * out, int rank, int nother, hipStream_t st) {
+  const int64_t span = pick_span_decl(nnz);
+  const int64_t nwaves = (nnz + span - 1) / span;
+  const int wpb = 4;
+  const int64_t nblocks = (nwaves + wpb - 1) / wpb;
+  dim3 grid((uint32_t)nblocks), block(wpb * WAVE);
+#define SARGS key, idx[0], idx[1], idx[2], idx[3], mats[0], mats[1], \
+              mats[2], mats[3], vals, nnz, span, out
+#define LS(F_, N_) \
+  hipLaunchKernelGGL((mttkrp_flat2_kern<V, F_, N_, 8, S>), grid, block, 0, \
+                     st, SARGS)
+#define LSF(N_) \
+  switch (rank) { case 4: LS(4, N_); break; case 8: LS(8, N_); break; \
+                  case 16: LS(16, N_); break; case 32: LS(32, N_); break; \
+                  default: LS(64, N_); break; }
+  switch (nother) {
+    case 2: LSF(2); break;
+    case 3: LSF(3); break;
+    default: LSF(4); break;
+  }
+#undef LSF
+#undef LS
+#undef SARGS
 }
 
 // ------------------------------------------- pipelined spec kernel (v3)
@@ -462,6 +498,8 @@ inline int64_t pick_span(int64_t nnz) {
   return span;
 }
 
+inline int64_t pick_span_decl(int64_t nnz) { return pick_span(nnz); }
+
 inline bool spec_ok(int F) {
   return F == 4 || F == 8 || F == 16 || F == 32 || F == 64;
 }
@@ -551,4 +589,27 @@ extern "C" void splatt_hip_mttkrp_flat_f32(
     void * stream) {
   launch_flat<float>(key, idx, mats, vals, nnz, out, rank, nother,
                      (hipStream_t)stream);
+}
+
+// reduced-precision factor STORAGE (f64 accumulate) on the v2 path;
+// returns nonzero for unsupported (non-spec) ranks
+extern "C" int splatt_hip_mttkrp_flat_f64f32(
+    const int32_t * key, const int32_t * const * idx,
+    const float * const * mats, const double * vals, int64_t nnz,
+    double * out, int rank, int nother, void * stream) {
+  if (!spec_ok(rank) || nother > 4) return -1;
+  launch_flat_store<double, float>(key, idx, mats, vals, nnz, out, rank,
+                                   nother, (hipStream_t)stream);
+  return 0;
+}
+
+extern "C" int splatt_hip_mttkrp_flat_f64bf16(
+    const int32_t * key, const int32_t * const * idx,
+    const uint16_t * const * mats, const double * vals, int64_t nnz,
+    double * out, int rank, int nother, void * stream) {
+  if (!spec_ok(rank) || nother > 4) return -1;
+  launch_flat_store<double, bf16>(key, idx,
+                                  (const bf16 * const *)mats, vals, nnz,
+                                  out, rank, nother, (hipStream_t)stream);
+  return 0;
 }

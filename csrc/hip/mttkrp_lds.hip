@@ -18,7 +18,12 @@
 #include <hip/hip_runtime.h>
 #include <cstdint>
 
+#include "store_types.hpp"
+
 namespace {
+using splatt_store::bf16;
+using splatt_store::VecElemT;
+using splatt_store::to_compute;
 
 constexpr int WAVE = 64;
 constexpr int WPB = 4;   // waves per block
@@ -155,25 +160,6 @@ mttkrp_flat5_kern(const int32_t * __restrict__ key,
 // S=bf16 with V=double is the documented reduced-precision factor-store
 // mode for HBM-bound shapes: gathered rows shrink 2-4x in cache lines
 // while every multiply-accumulate stays f64 (ROADMAP item 2b).
-struct bf16 { uint16_t v; };
-template <typename S> struct VecElemT { using type = S; };
-template <> struct VecElemT<bf16> { using type = uint16_t; };
-__device__ __forceinline__ double to_compute(double x, double) { return x; }
-__device__ __forceinline__ float to_compute(float x, float) { return x; }
-__device__ __forceinline__ double to_compute(float x, double) {
-  return (double)x;
-}
-__device__ __forceinline__ double to_compute(bf16 x, double) {
-  union { uint32_t u; float f; } c;
-  c.u = (uint32_t)x.v << 16;
-  return (double)c.f;
-}
-__device__ __forceinline__ float to_compute(bf16 x, float) {
-  union { uint32_t u; float f; } c;
-  c.u = (uint32_t)x.v << 16;
-  return c.f;
-}
-
 template <typename V, int F, int NOTHER, typename S = V>
 __global__ void __launch_bounds__(WPB * WAVE)
 mttkrp_flat6_kern(const int * __restrict__ pack_raw,

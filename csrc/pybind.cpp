@@ -433,6 +433,12 @@ void splatt_hip_mttkrp_flat6_f32(
     const int32_t*, const float*, const float*, const float*,
     const float*, const int64_t*, const int64_t*, const int32_t*, int64_t,
     int32_t, int32_t, float*, int, int, void*);
+int splatt_hip_mttkrp_flat_f64f32(
+    const int32_t*, const int32_t* const*, const float* const*,
+    const double*, int64_t, double*, int, int, void*);
+int splatt_hip_mttkrp_flat_f64bf16(
+    const int32_t*, const int32_t* const*, const uint16_t* const*,
+    const double*, int64_t, double*, int, int, void*);
 void splatt_hip_mttkrp_flat6_f64f32(
     const int32_t*, const float*, const float*, const float*,
     const double*, const int64_t*, const int64_t*, const int32_t*, int64_t,
@@ -575,6 +581,29 @@ static void py_gpu_mttkrp_flat(Tensor key, std::vector<Tensor> idx,
   const int32_t * ip[8] = {};
   for (int t = 0; t < nother; ++t) ip[t] = idx[t].data_ptr<int32_t>();
   if (vals.scalar_type() == torch::kFloat64) {
+    // reduced-precision factor STORAGE with f64 accumulation (v2 path)
+    if (mats[0].scalar_type() == torch::kFloat32) {
+      const float * mp[8] = {};
+      for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+      TORCH_CHECK(splatt_hip_mttkrp_flat_f64f32(
+                      key.data_ptr<int32_t>(), ip, mp,
+                      vals.data_ptr<double>(), nnz, out.data_ptr<double>(),
+                      rank, nother, (void*)stream) == 0,
+                  "f32 factor store needs rank in {4,8,16,32,64}");
+      return;
+    }
+    if (mats[0].scalar_type() == torch::kBFloat16) {
+      const uint16_t * mp[8] = {};
+      for (int t = 0; t < nother; ++t)
+        mp[t] = reinterpret_cast<const uint16_t*>(
+            mats[t].data_ptr<at::BFloat16>());
+      TORCH_CHECK(splatt_hip_mttkrp_flat_f64bf16(
+                      key.data_ptr<int32_t>(), ip, mp,
+                      vals.data_ptr<double>(), nnz, out.data_ptr<double>(),
+                      rank, nother, (void*)stream) == 0,
+                  "bf16 factor store needs rank in {4,8,16,32,64}");
+      return;
+    }
     const double * mp[8] = {};
     for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
     splatt_hip_mttkrp_flat_f64(key.data_ptr<int32_t>(), ip, mp,

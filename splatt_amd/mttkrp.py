@@ -324,10 +324,10 @@ def mttkrp(src: CsfSet | Csf, mats: List[torch.Tensor], mode: int,
             raise ValueError(
                 f"factor {m}: dtype/device {A.dtype}/{A.device} does not "
                 f"match tensor {c.vals.dtype}/{c.device}")
-    if storage and getattr(c, "_pack", None) is None:
+    if storage and rank not in (4, 8, 16, 32, 64):
         raise ValueError(
-            "reduced-precision factor storage needs an LDS-staged packed "
-            "build (build with stage_rank=rank; see SPLATT_FACTOR_STORE)")
+            "reduced-precision factor storage (SPLATT_FACTOR_STORE) needs "
+            f"rank in {{4,8,16,32,64}}, got {rank}")
     if out is None:
         out = torch.empty(c.dims[mode], rank, dtype=c.vals.dtype,
                           device=mats[0].device)

@@ -517,18 +517,21 @@ def test_gpu_packed_stream_matches(t3, rank):
             assert (out6.cpu() - ref).abs().max().item() < 1e-8
 
 
+@pytest.mark.parametrize("staged", [True, False])
 @pytest.mark.parametrize("store", ["f32", "bf16"])
-def test_gpu_factor_store_mttkrp(t3, store):
+def test_gpu_factor_store_mttkrp(t3, store, staged):
     """Reduced-precision factor STORAGE (SPLATT_FACTOR_STORE): gathers
     read f32/bf16 rows, accumulation stays f64; result tracks the f64
-    oracle within storage precision (ROADMAP 2b documented mode)."""
+    oracle within storage precision (ROADMAP 2b documented mode), on
+    both the LDS-staged packed path and the plain v2 path."""
     import os
     from splatt_amd.parallel.dist_cpd import build_shard_csf
     rank = 16
     os.environ["SPLATT_FACTOR_STORE"] = store
     try:
         cs = build_shard_csf(t3.to("cuda"), list(t3.dims), "all",
-                             flat_only=True, stage_rank=rank)
+                             flat_only=True,
+                             stage_rank=rank if staged else 0)
         mats_c = make_mats(t3.dims, rank)
         qdt = {"f32": torch.float32, "bf16": torch.bfloat16}[store]
         mats_q = [m.cuda().to(qdt) for m in mats_c]
