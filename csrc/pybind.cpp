@@ -489,6 +489,59 @@ static void py_gpu_mttkrp_flat5(Tensor key, std::vector<Tensor> idx,
   }
 }
 
+extern "C" {
+int splatt_hip_mttkrp_det6_f64(
+    const int32_t*, const double*, const double*, const double*,
+    const double*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    int32_t, int32_t, int64_t, int64_t, double*, double*, double*, int,
+    int, void*);
+int splatt_hip_mttkrp_det6_f32(
+    const int32_t*, const float*, const float*, const float*,
+    const float*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    int32_t, int32_t, int64_t, int64_t, float*, float*, float*, int, int,
+    void*);
+}
+
+// LDS-staged bitwise-deterministic MTTKRP over the packed stream:
+// per-bucket privatized outputs + ordered fixup + ascending-bucket fold
+static void py_gpu_mttkrp_det6(Tensor pack, std::vector<Tensor> mats,
+                               Tensor vals, Tensor blk_start,
+                               Tensor blk_end, Tensor blk_row0,
+                               int64_t chunk, int64_t dim0,
+                               int64_t nbuckets, Tensor outb, Tensor side,
+                               Tensor out, int64_t stream) {
+  const int nother = (int)mats.size();
+  TORCH_CHECK(nother >= 2 && nother <= 3);
+  TORCH_CHECK(pack.is_contiguous() && pack.size(1) == 4
+              && pack.scalar_type() == torch::kInt32);
+  const int rank = (int)mats[0].size(1);
+  const int64_t nblocks = blk_start.numel();
+  const int64_t nrows_out = out.size(0);
+  int rc;
+  if (vals.scalar_type() == torch::kFloat64) {
+    const double * mp[3] = {nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<double>();
+    rc = splatt_hip_mttkrp_det6_f64(
+        pack.data_ptr<int32_t>(), mp[0], mp[1], mp[2],
+        vals.data_ptr<double>(), blk_start.data_ptr<int64_t>(),
+        blk_end.data_ptr<int64_t>(), blk_row0.data_ptr<int32_t>(), nblocks,
+        (int32_t)chunk, (int32_t)dim0, nrows_out, nbuckets,
+        outb.data_ptr<double>(), side.data_ptr<double>(),
+        out.data_ptr<double>(), rank, nother, (void*)stream);
+  } else {
+    const float * mp[3] = {nullptr, nullptr, nullptr};
+    for (int t = 0; t < nother; ++t) mp[t] = mats[t].data_ptr<float>();
+    rc = splatt_hip_mttkrp_det6_f32(
+        pack.data_ptr<int32_t>(), mp[0], mp[1], mp[2],
+        vals.data_ptr<float>(), blk_start.data_ptr<int64_t>(),
+        blk_end.data_ptr<int64_t>(), blk_row0.data_ptr<int32_t>(), nblocks,
+        (int32_t)chunk, (int32_t)dim0, nrows_out, nbuckets,
+        outb.data_ptr<float>(), side.data_ptr<float>(),
+        out.data_ptr<float>(), rank, nother, (void*)stream);
+  }
+  TORCH_CHECK(rc == 0, "det6 supports spec ranks and 3-4 modes, rc=", rc);
+}
+
 // packed-stream LDS variant: one int4 word per nonzero
 // (x = output key, y = staged level, z/w = remaining levels)
 static void py_gpu_mttkrp_flat6(Tensor pack, std::vector<Tensor> mats,
@@ -764,6 +817,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "LDS-staged flat MTTKRP (bucketed builds, root output)");
   m.def("gpu_mttkrp_flat6", &py_gpu_mttkrp_flat6,
         "LDS-staged flat MTTKRP over the packed int4 stream");
+  m.def("gpu_mttkrp_det6", &py_gpu_mttkrp_det6,
+        "LDS-staged deterministic MTTKRP (bucket-privatized outputs)");
   m.def("partition_weighted", [](std::vector<int64_t> w, int nparts) {
     int64_t bn = 0;
     auto parts = sp::partition_weighted(w.data(), (int64_t)w.size(), nparts, &bn);

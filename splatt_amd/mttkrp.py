@@ -159,6 +159,39 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
         raise ValueError(
             f"deterministic MTTKRP supports ranks 4/8/16/32/64 and <=5 "
             f"modes, got rank {rank}, {nm} modes")
+    # LDS-staged deterministic kernel (det6): same packed bucket-major
+    # stream as the default v6 path, output privatized per bucket, ordered
+    # fixup + ascending-bucket fold (csrc/hip/mttkrp_det.hip). Used when
+    # the bucket-partial workspace fits SPLATT_DET_MB (default 4096 MB).
+    pack = getattr(c, "_pack", None)
+    st6 = getattr(c, "_stage", None)
+    if (pack is not None and st6 is not None and nm <= 4
+            and os.environ.get("SPLATT_NO_DET6") != "1"):
+        blocks = _stage_blocks(c)
+        nbuckets = st6["nbuckets"]
+        nrows = c.dims[c.dim_perm[0]]
+        budget = int(os.environ.get("SPLATT_DET_MB", "4096")) << 20
+        need_outb = nbuckets * nrows * rank * c.vals.element_size()
+        if need_outb <= budget:
+            nblocks = int(blocks["start"].numel())
+            nwalk = nblocks * 4 * (64 // rank)
+            ws = getattr(c, "_det6_ws", None)
+            if (ws is None or ws[0].numel() < nbuckets * nrows * rank
+                    or ws[1].numel() < nwalk * 2 * rank):
+                outb = torch.empty(nbuckets * nrows * rank,
+                                   dtype=c.vals.dtype, device=c.device)
+                side = torch.empty(nwalk * 2 * rank, dtype=c.vals.dtype,
+                                   device=c.device)
+                ws = (outb, side)
+                object.__setattr__(c, "_det6_ws", ws)
+            levels = c._pack_levels  # type: ignore[attr-defined]
+            ms = [mats[c.dim_perm[l]].contiguous() for l in levels[1:]]
+            stream = torch.cuda.current_stream().cuda_stream
+            native().gpu_mttkrp_det6(
+                pack, ms, c.vals, blocks["start"], blocks["end"],
+                blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[0]],
+                nbuckets, ws[0], ws[1], out, stream)
+            return
     # the deterministic scheme needs every output key CONTIGUOUS in the
     # stream. LDS-bucketed builds reorder the stream bucket-major, so a
     # key-sorted copy is built once and cached (stable argsort -> the
