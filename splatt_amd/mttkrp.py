@@ -189,7 +189,8 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
             stream = torch.cuda.current_stream().cuda_stream
             native().gpu_mttkrp_det6(
                 pack, ms, c.vals, blocks["start"], blocks["end"],
-                blocks["row0"], blocks["chunk"], c.dims[c.dim_perm[0]],
+                blocks["row0"], blocks["chunk"],
+                c.dims[c.dim_perm[st6["level"]]],   # staged level's dim
                 nbuckets, ws[0], ws[1], out, stream)
             return
     # the deterministic scheme needs every output key CONTIGUOUS in the
