@@ -42,10 +42,17 @@ def spd_inverse(G: torch.Tensor) -> torch.Tensor:
         native().gpu_spd_inverse(G.contiguous(), Ginv,
                                  torch.cuda.current_stream().cuda_stream)
         return Ginv
-    L = torch.linalg.cholesky(
-        G + 1e-12 * G.diagonal().abs().max()
-        * torch.eye(F, dtype=G.dtype, device=G.device))
-    return torch.cholesky_inverse(L)
+    eye = torch.eye(F, dtype=G.dtype, device=G.device)
+    scale = float(G.diagonal().abs().max().clamp(min=1.0))
+    jitter = 1e-12
+    for _ in range(12):   # escalating Tikhonov (reference gelss fallback
+        try:              # role, matrix.c:554-599)
+            L = torch.linalg.cholesky(G + jitter * scale * eye)
+            return torch.cholesky_inverse(L)
+        except torch.linalg.LinAlgError:
+            jitter *= 100.0
+    # last resort: pseudo-inverse (rank-deficient Gram)
+    return torch.linalg.pinv(G)
 
 
 def solve_rows(mb: torch.Tensor, Ginv: torch.Tensor) -> torch.Tensor:
