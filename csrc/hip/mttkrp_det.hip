@@ -306,6 +306,7 @@ mttkrp_det6_fixup_kern(const int * __restrict__ pack_raw,
                        const int64_t * __restrict__ blk_start,
                        const int64_t * __restrict__ blk_end,
                        const int32_t * __restrict__ blk_row0,
+                       const int64_t * __restrict__ blk_bucket_p0,
                        int64_t nblocks, int32_t chunk, int64_t nrows_out,
                        const V * __restrict__ side, V * __restrict__ outb) {
   using Pack = __attribute__((ext_vector_type(4))) int;
@@ -359,21 +360,17 @@ mttkrp_det6_fixup_kern(const int * __restrict__ pack_raw,
   // unique first contributor of kf within this BUCKET: either p0 is the
   // bucket's first stream position (pack[p0-1] belongs to another
   // bucket and may coincidentally equal kf), or the previous element has
-  // a different key. blk arrays are emitted bucket-major, so the
-  // bucket's first position is blk_start of its first block.
-  bool is_first;
-  if (p0 == 0) {
-    is_first = true;
-  } else {
-    int64_t fb = w / NSUB;
-    while (fb > 0 && blk_row0[fb - 1] == myrow0) --fb;
-    is_first = (p0 == blk_start[fb]) || (pack[p0 - 1].x != kf);
-  }
+  // a different key. The bucket's first position travels with the block
+  // table (host-precomputed, splatt_amd/mttkrp.py _stage_blocks).
+  const bool is_first = (p0 == blk_bucket_p0[w / NSUB])
+                        || (pack[p0 - 1].x != kf);
   if (is_first) resolve(kf);
   if (kl != kf) resolve(kl);
 }
 
-// fold bucket slices in ascending bucket order (deterministic)
+// fold bucket slices in ascending bucket order (deterministic; the
+// 4-deep bucket unroll keeps enough loads in flight per thread — the
+// element count alone is too small to cover DRAM latency)
 template <typename V>
 __global__ void __launch_bounds__(256)
 det6_fold_kern(const V * __restrict__ outb, int64_t nbuckets,
@@ -381,10 +378,16 @@ det6_fold_kern(const V * __restrict__ outb, int64_t nbuckets,
   const int64_t t0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t e = t0; e < elems; e += stride) {
-    V s = (V)0;
-    for (int64_t b = 0; b < nbuckets; ++b)
-      s += outb[b * elems + e];
-    out[e] = s;
+    V s0 = (V)0, s1 = (V)0, s2 = (V)0, s3 = (V)0;
+    int64_t b = 0;
+    for (; b + 4 <= nbuckets; b += 4) {
+      s0 += outb[b * elems + e];
+      s1 += outb[(b + 1) * elems + e];
+      s2 += outb[(b + 2) * elems + e];
+      s3 += outb[(b + 3) * elems + e];
+    }
+    for (; b < nbuckets; ++b) s0 += outb[b * elems + e];
+    out[e] = ((s0 + s1) + (s2 + s3));
   }
 }
 
@@ -392,6 +395,7 @@ template <typename V>
 int launch_det6(const int32_t * pack, const V * const mats[3],
                 const V * vals, const int64_t * blk_start,
                 const int64_t * blk_end, const int32_t * blk_row0,
+                const int64_t * blk_bucket_p0,
                 int64_t nblocks, int32_t chunk, int32_t dim0,
                 int64_t nrows_out, int64_t nbuckets, V * outb, V * side,
                 V * out, int rank, int nother, hipStream_t st) {
@@ -413,7 +417,7 @@ int launch_det6(const int32_t * pack, const V * const mats[3],
     const int64_t fb = (nw + tpb / F_ - 1) / (tpb / F_); \
     hipLaunchKernelGGL((mttkrp_det6_fixup_kern<V, F_>), \
         dim3((uint32_t)fb), dim3(tpb), 0, st, pack, blk_start, blk_end, \
-        blk_row0, nblocks, chunk, nrows_out, side, outb); }
+        blk_row0, blk_bucket_p0, nblocks, chunk, nrows_out, side, outb); }
 #define D6F(N_) \
   switch (rank) { case 4: D6(4, N_); break; case 8: D6(8, N_); break; \
                   case 16: D6(16, N_); break; case 32: D6(32, N_); break; \
@@ -493,13 +497,14 @@ int launch_flat_det(const int32_t * key, const int32_t * const idx[8],
 extern "C" int splatt_hip_mttkrp_det6_f64(
     const int32_t * pack, const double * m0, const double * m1,
     const double * m2, const double * vals, const int64_t * blk_start,
-    const int64_t * blk_end, const int32_t * blk_row0, int64_t nblocks,
+    const int64_t * blk_end, const int32_t * blk_row0,
+    const int64_t * blk_bucket_p0, int64_t nblocks,
     int32_t chunk, int32_t dim0, int64_t nrows_out, int64_t nbuckets,
     double * outb, double * side, double * out, int rank, int nother,
     void * stream) {
   const double * mats[3] = {m0, m1, m2};
   return launch_det6<double>(pack, mats, vals, blk_start, blk_end,
-                             blk_row0, nblocks, chunk, dim0, nrows_out,
+                             blk_row0, blk_bucket_p0, nblocks, chunk, dim0, nrows_out,
                              nbuckets, outb, side, out, rank, nother,
                              (hipStream_t)stream);
 }
@@ -507,15 +512,16 @@ extern "C" int splatt_hip_mttkrp_det6_f64(
 extern "C" int splatt_hip_mttkrp_det6_f32(
     const int32_t * pack, const float * m0, const float * m1,
     const float * m2, const float * vals, const int64_t * blk_start,
-    const int64_t * blk_end, const int32_t * blk_row0, int64_t nblocks,
+    const int64_t * blk_end, const int32_t * blk_row0,
+    const int64_t * blk_bucket_p0, int64_t nblocks,
     int32_t chunk, int32_t dim0, int64_t nrows_out, int64_t nbuckets,
     float * outb, float * side, float * out, int rank, int nother,
     void * stream) {
   const float * mats[3] = {m0, m1, m2};
   return launch_det6<float>(pack, mats, vals, blk_start, blk_end,
-                            blk_row0, nblocks, chunk, dim0, nrows_out,
-                            nbuckets, outb, side, out, rank, nother,
-                            (hipStream_t)stream);
+                            blk_row0, blk_bucket_p0, nblocks, chunk, dim0,
+                            nrows_out, nbuckets, outb, side, out, rank,
+                            nother, (hipStream_t)stream);
 }
 
 extern "C" int64_t splatt_hip_flat_det_ws(int64_t nnz, int rank) {

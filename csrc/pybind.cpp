@@ -492,12 +492,14 @@ static void py_gpu_mttkrp_flat5(Tensor key, std::vector<Tensor> idx,
 extern "C" {
 int splatt_hip_mttkrp_det6_f64(
     const int32_t*, const double*, const double*, const double*,
-    const double*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    const double*, const int64_t*, const int64_t*, const int32_t*,
+    const int64_t*, int64_t,
     int32_t, int32_t, int64_t, int64_t, double*, double*, double*, int,
     int, void*);
 int splatt_hip_mttkrp_det6_f32(
     const int32_t*, const float*, const float*, const float*,
-    const float*, const int64_t*, const int64_t*, const int32_t*, int64_t,
+    const float*, const int64_t*, const int64_t*, const int32_t*,
+    const int64_t*, int64_t,
     int32_t, int32_t, int64_t, int64_t, float*, float*, float*, int, int,
     void*);
 }
@@ -507,6 +509,7 @@ int splatt_hip_mttkrp_det6_f32(
 static void py_gpu_mttkrp_det6(Tensor pack, std::vector<Tensor> mats,
                                Tensor vals, Tensor blk_start,
                                Tensor blk_end, Tensor blk_row0,
+                               Tensor blk_bucket_p0,
                                int64_t chunk, int64_t dim0,
                                int64_t nbuckets, Tensor outb, Tensor side,
                                Tensor out, int64_t stream) {
@@ -524,7 +527,8 @@ static void py_gpu_mttkrp_det6(Tensor pack, std::vector<Tensor> mats,
     rc = splatt_hip_mttkrp_det6_f64(
         pack.data_ptr<int32_t>(), mp[0], mp[1], mp[2],
         vals.data_ptr<double>(), blk_start.data_ptr<int64_t>(),
-        blk_end.data_ptr<int64_t>(), blk_row0.data_ptr<int32_t>(), nblocks,
+        blk_end.data_ptr<int64_t>(), blk_row0.data_ptr<int32_t>(),
+        blk_bucket_p0.data_ptr<int64_t>(), nblocks,
         (int32_t)chunk, (int32_t)dim0, nrows_out, nbuckets,
         outb.data_ptr<double>(), side.data_ptr<double>(),
         out.data_ptr<double>(), rank, nother, (void*)stream);
@@ -534,7 +538,8 @@ static void py_gpu_mttkrp_det6(Tensor pack, std::vector<Tensor> mats,
     rc = splatt_hip_mttkrp_det6_f32(
         pack.data_ptr<int32_t>(), mp[0], mp[1], mp[2],
         vals.data_ptr<float>(), blk_start.data_ptr<int64_t>(),
-        blk_end.data_ptr<int64_t>(), blk_row0.data_ptr<int32_t>(), nblocks,
+        blk_end.data_ptr<int64_t>(), blk_row0.data_ptr<int32_t>(),
+        blk_bucket_p0.data_ptr<int64_t>(), nblocks,
         (int32_t)chunk, (int32_t)dim0, nrows_out, nbuckets,
         outb.data_ptr<float>(), side.data_ptr<float>(),
         out.data_ptr<float>(), rank, nother, (void*)stream);

@@ -91,7 +91,7 @@ def _stage_blocks(c: Csf, rows: tuple | None = None) -> dict:
             if p0 < p1:
                 segs.append((p0, p1, b))
     tgt = max(4096, nnz // int(os.environ.get("SPLATT_LDS_BLOCKS", "16384")))
-    starts, ends, row0s = [], [], []
+    starts, ends, row0s, bkt0s = [], [], [], []
     for s, e, b in segs:
         p = s
         while p < e:
@@ -99,11 +99,13 @@ def _stage_blocks(c: Csf, rows: tuple | None = None) -> dict:
             starts.append(p)
             ends.append(q)
             row0s.append(b * chunk)
+            bkt0s.append(s)       # bucket's first stream position
             p = q
     blocks = {
         "start": torch.tensor(starts, dtype=torch.int64, device=dev),
         "end": torch.tensor(ends, dtype=torch.int64, device=dev),
         "row0": torch.tensor(row0s, dtype=torch.int32, device=dev),
+        "bucket_p0": torch.tensor(bkt0s, dtype=torch.int64, device=dev),
         "chunk": chunk,
         "level": lvl,
     }
@@ -189,7 +191,7 @@ def _gpu_mttkrp_det(c: Csf, depth: int, mats: List[torch.Tensor],
             stream = torch.cuda.current_stream().cuda_stream
             native().gpu_mttkrp_det6(
                 pack, ms, c.vals, blocks["start"], blocks["end"],
-                blocks["row0"], blocks["chunk"],
+                blocks["row0"], blocks["bucket_p0"], blocks["chunk"],
                 c.dims[c.dim_perm[st6["level"]]],   # staged level's dim
                 nbuckets, ws[0], ws[1], out, stream)
             return
