@@ -25,14 +25,12 @@ if pmc in tables:
     cols = [r[1] for r in db.execute(f"PRAGMA table_info({pmc})")]
     print("\n# PMC table columns:", cols)
     q2 = f"""
-    SELECT ks.display_name, s.string, SUM(p.value), COUNT(*)
+    SELECT ks.display_name, i.name, SUM(p.value), COUNT(*)
     FROM {pmc} p
     JOIN rocpd_info_pmc_{sfx} i ON p.pmc_id = i.id
-    JOIN rocpd_string_{sfx} s ON i.name_id = s.id
-    JOIN rocpd_event_{sfx} e ON p.event_id = e.id
-    JOIN rocpd_kernel_dispatch_{sfx} k ON k.event_id = e.id
+    JOIN rocpd_kernel_dispatch_{sfx} k ON k.event_id = p.event_id
     JOIN rocpd_info_kernel_symbol_{sfx} ks ON k.kernel_id = ks.id
-    GROUP BY ks.display_name, s.string
+    GROUP BY ks.display_name, i.name
     HAVING SUM(p.value) > 0 ORDER BY 1, 2
     """
     try:
