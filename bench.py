@@ -164,6 +164,9 @@ def main():
                          gather_tiles=gt if device.type == "cuda" else 0,
                          stage_rank=stage_rank)
     del shard
+    # actual staging outcome (the run-density gate may decline staging,
+    # e.g. Amazon/Delicious-scale dims -> plain v2 kernels)
+    staged = any(getattr(c, "_stage", None) is not None for c in cs.csfs)
     if rank == 0:
         print(f"# setup: grid={dec.grid} shard_nnz={nnz_local} gt={gt} "
               f"csf_bytes={cs.storage_bytes()} build_s={time.time() - t0:.1f}",
@@ -259,7 +262,7 @@ def main():
                 "nnz": nnz_global,
                 "cp_rank": rank_f,
                 "csf": args.csf,
-                "tiling": ("lds-staged buckets" if stage_rank else
+                "tiling": ("lds-staged buckets" if staged else
                            (f"gather-range x{gt}" if gt > 1 else "none")),
                 "parallelism": f"{args.decomp}-grid {dec.grid} x{world} (RCCL/xGMI)",
                 "fit": round(st.fit, 6),

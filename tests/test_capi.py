@@ -255,3 +255,19 @@ def test_capi_cpd_uses_gpu_engine(lib, tmp_path):
     lib.splatt_free_kruskal(ctypes.byref(k2))
     lib.splatt_free_csf(csf, o)
     lib.splatt_free_opts(o)
+
+
+def test_mex_sources_compile(tmp_path):
+    """The Octave/MATLAB MEX bindings (matlab/*.c) must be valid C
+    against the public header — compile-checked with a stub mex.h since
+    no MEX toolchain ships in this image (matlab/README.md)."""
+    import glob
+    srcs = sorted(glob.glob(os.path.join(ROOT, "matlab", "*.c")))
+    assert len(srcs) >= 4
+    for src in srcs:
+        r = subprocess.run(
+            ["gcc", "-fsyntax-only", "-Wall", "-Werror",
+             "-I" + os.path.join(ROOT, "tests", "mex_stub"),
+             "-I" + os.path.join(ROOT, "csrc", "capi"), src],
+            capture_output=True, text=True)
+        assert r.returncode == 0, (src, r.stderr[-800:])

@@ -114,8 +114,9 @@ def test_gpu_spd_inverse_near_singular():
     G = (A.T @ A).cuda()
     Ginv = spd_inverse(G)
     assert torch.isfinite(Ginv).all()
-    x = torch.rand(5, F, dtype=torch.float64).cuda()
-    assert float(((x @ G) @ Ginv - x).abs().max()) < 1e-2
+    # regularized-inverse criterion (collinear subspace unrecoverable):
+    r = G @ Ginv @ G - G
+    assert float(r.abs().max() / G.abs().max()) < 1e-3
 
 
 @pytest.mark.gpu
