@@ -45,6 +45,8 @@ CONFIGS = {
     "amazon": ([4_821_207, 1_774_269, 1_805_187], 1_741_809_018, 16, 1.5),
     "delicious4d": ([532_924, 17_262_471, 2_480_308, 1443], 140_126_181, 32, 1.5),
     "small": ([1200, 900, 1500], 300_000, 16, 1.5),
+    # CI-sized 4-mode shape (config 5's mode count at gloo scale)
+    "small4": ([500, 1400, 700, 60], 250_000, 8, 1.5),
 }
 
 

@@ -75,3 +75,22 @@ def test_bench_driver_argv_world8(tmp_path):
     assert j["scaling"] == "strong"
     assert "medium-grid" in j["config"]["parallelism"]
     assert 0.0 <= j["config"]["fit"] < 1.0
+
+
+@pytest.mark.timeout(600)
+def test_bench_driver_argv_world8_4mode(tmp_path):
+    """Config 5's SHAPE class (4-mode, medium-grain, N=8) through the
+    driver's exact argv over gloo + forced RS/AG primitives."""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               SPLATT_BENCH_CONFIG="small4",
+               SPLATT_FORCE_RS_PRIMS="1")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--nnodes=1", "--nproc-per-node", "8", "--local-addr", "127.0.0.1",
+         "bench.py", "--gpus", "8", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, cwd=ROOT, env=env, timeout=580)
+    assert r.returncode == 0, r.stderr[-1500:]
+    j = json.loads([l for l in r.stdout.strip().splitlines()
+                    if l.startswith("{")][0])
+    assert j["n_gpus"] == 8 and j["scaling"] == "strong"
+    assert len(j["config"]["dims"]) == 4
