@@ -104,7 +104,7 @@ def main():
                          "numbers are full f64")
     ap.add_argument("--deterministic", action="store_true",
                     help="bitwise-reproducible kernels (SPLATT_DETERMINISTIC=1;"
-                         " ~60%% of default throughput)")
+                         " ~84%% of default throughput via the det6 kernel)")
     ap.add_argument("--decomp",
                     default=os.environ.get("SPLATT_BENCH_DECOMP", "medium"),
                     choices=["coarse", "medium"],
