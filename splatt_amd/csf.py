@@ -86,9 +86,19 @@ class Csf:
 
     def storage_bytes(self) -> int:
         b = self.vals.numel() * self.vals.element_size()
+        pack = getattr(self, "_pack", None)
+        if pack is not None:
+            # expansions/leaf fids are strided views into the pack
+            return b + pack.numel() * pack.element_size()
         for t in self.fptr + self.fids:
             if t is not None:
                 b += t.numel() * t.element_size()
+        # flat builds keep contiguous expansions instead of the tree
+        cache = getattr(self, "_expand_cache", None)
+        if cache and all(fp is None for fp in self.fptr):
+            for l, t in cache.items():
+                if t is not self.fids[self.nmodes - 1]:
+                    b += t.numel() * t.element_size()
         return b
 
     def ancestor_expand(self, level: int) -> torch.Tensor:
