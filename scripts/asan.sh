@@ -5,8 +5,18 @@ set -euo pipefail
 cd "$(dirname "$0")/.."
 g++ -std=c++17 -O1 -g -fopenmp -fsanitize=address,undefined \
     -fno-omit-frame-pointer -Icsrc \
-    csrc/core/*.cpp csrc/capi/capi.cpp csrc/capi/splatt_main.cpp \
+    -D__HIP_PLATFORM_AMD__=1 -I"${ROCM_PATH:-/opt/rocm}/include" \
+    csrc/core/*.cpp csrc/capi/capi.cpp csrc/capi/capi_gpu.cpp \
+    csrc/capi/splatt_main.cpp \
+    build/hip_obj/dense_kernels.o build/hip_obj/mttkrp_det.o \
+    build/hip_obj/mttkrp_flat.o build/hip_obj/mttkrp_kernels.o \
+    build/hip_obj/mttkrp_lds.o \
+    -L"${ROCM_PATH:-/opt/rocm}/lib" -lamdhip64 \
+    -Wl,-rpath,"${ROCM_PATH:-/opt/rocm}/lib" \
     -o /tmp/splatt_asan
+# (the HIP kernel objects are linked un-sanitized; on a GPU-less box
+# capi_gpu_available() is false and the sanitized CPU core serves every
+# call)
 python - <<'PY'
 import splatt_amd as sp
 sp.SpTensor.synthetic([80, 60, 100], 20000, seed=11).save('/tmp/asan_t.tns')
