@@ -134,7 +134,13 @@ def main():
     if world > 1:
         if args.device != "cpu":
             torch.cuda.set_device(local_rank % torch.cuda.device_count())
-        dist.init_process_group("nccl" if args.device != "cpu" else "gloo")
+        # SPLATT_BENCH_BACKEND=gloo lets CI drive CUDA tensors through
+        # gloo with several ranks sharing one GPU (chunked-pipeline
+        # composition test); the driver's real runs use RCCL
+        backend = os.environ.get(
+            "SPLATT_BENCH_BACKEND",
+            "nccl" if args.device != "cpu" else "gloo")
+        dist.init_process_group(backend)
     device = torch.device(args.device if args.device == "cpu" else
                           f"cuda:{local_rank % torch.cuda.device_count()}")
 
