@@ -17,6 +17,7 @@
 // depths fall back to the v2 kernel.
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 #include "store_types.hpp"
 
@@ -160,7 +161,7 @@ mttkrp_flat5_kern(const int32_t * __restrict__ key,
 // S=bf16 with V=double is the documented reduced-precision factor-store
 // mode for HBM-bound shapes: gathered rows shrink 2-4x in cache lines
 // while every multiply-accumulate stays f64 (ROADMAP item 2b).
-template <typename V, int F, int NOTHER, typename S = V>
+template <typename V, int F, int NOTHER, typename S = V, int GBP = 8>
 __global__ void __launch_bounds__(WPB * WAVE)
 mttkrp_flat6_kern(const int * __restrict__ pack_raw,
                   const S * __restrict__ m0, const S * __restrict__ m1,
@@ -173,7 +174,7 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
                   V * __restrict__ out) {
   using Pack = __attribute__((ext_vector_type(4))) int;
   const Pack * __restrict__ pack = reinterpret_cast<const Pack *>(pack_raw);
-  constexpr int GB = 8;
+  constexpr int GB = (F >= GBP) ? GBP : F;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   S * smem = reinterpret_cast<S *>(smem_raw);
 
@@ -270,10 +271,18 @@ void launch_flat6(const int32_t * pack, const S * const mats[3],
                   int rank, int nother, hipStream_t st) {
   dim3 grid((uint32_t)nblocks), block(WPB * WAVE);
   const size_t lds = (size_t)chunk * rank * sizeof(S);
+  // gather-batch depth A/B lever (register pressure vs loads in flight)
+  const char * ge = getenv("SPLATT_V6_GB");
+  const int gb = ge ? atoi(ge) : 8;
 #define ARGS6 pack, mats[0], mats[1], mats[2], vals, blk_start, blk_end, \
               blk_row0, chunk, dim0, out
 #define L6(F_, N_) \
-  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S>), grid, block, lds, st, ARGS6)
+  switch (gb) { \
+    case 4:  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 4>), grid, block, lds, st, ARGS6); break; \
+    case 6:  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 6>), grid, block, lds, st, ARGS6); break; \
+    case 12: hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 12>), grid, block, lds, st, ARGS6); break; \
+    case 16: hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 16>), grid, block, lds, st, ARGS6); break; \
+    default: hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 8>), grid, block, lds, st, ARGS6); break; }
 #define L6F(N_) \
   switch (rank) { case 4: L6(4, N_); break; case 8: L6(8, N_); break; \
                   case 16: L6(16, N_); break; case 32: L6(32, N_); break; \
