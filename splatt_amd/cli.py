@@ -30,6 +30,8 @@ def _load(args) -> "sp.SpTensor":
 
 def cmd_cpd(args) -> int:
     import os
+    if getattr(args, "factor_store", "f64") != "f64":
+        os.environ["SPLATT_FACTOR_STORE"] = args.factor_store
     if getattr(args, "deterministic", False):
         # bitwise-reproducible device kernels (docs/KERNELS.md); forces
         # ALLMODE so every mode has a root-sorted stream
@@ -222,6 +224,11 @@ def main(argv=None) -> int:
     p.add_argument("--native", action="store_true",
                    help="use the C++ host driver (CPU reference path)")
     p.add_argument("--nowrite", action="store_true")
+    p.add_argument("--factor-store", default="f64",
+                   choices=["f64", "f32", "bf16"],
+                   help="reduced-precision factor STORAGE for the device "
+                        "MTTKRP gathers (accumulation stays f64; "
+                        "docs/TUNING.md)")
     p.add_argument("--deterministic", action="store_true",
                    help="bitwise-reproducible device CPD (atomic-free "
                         "kernels, ~60%% throughput; implies --csf all)")
