@@ -26,17 +26,22 @@ def one(i):
     stage = rng.choice([0, rank]) if flat else 0
     cfg = dict(i=i, dims=dims, nnz=nnz, rank=rank, policy=policy,
                flat=flat, stage=stage)
-    t = sp.SpTensor.synthetic(dims, nnz, seed=SEED + i).fixed(dedup=True)
-    mats_c = [sp.seeded_init(d, rank, m, 5 + i) for m, d in enumerate(dims)]
+    dtype = rng.choice([torch.float64, torch.float64, torch.float32])
+    cfg["dtype"] = str(dtype)
+    t = sp.SpTensor.synthetic(dims, nnz, seed=SEED + i,
+                              dtype=dtype).fixed(dedup=True)
+    mats_c = [sp.seeded_init(d, rank, m, 5 + i, dtype=dtype)
+              for m, d in enumerate(dims)]
     mats_g = [m.cuda() for m in mats_c]
     cs = build_shard_csf(t.to("cuda"), dims, policy, flat_only=flat,
                          gather_tiles=0, stage_rank=stage)
+    tol = 1e-8 if dtype == torch.float64 else 1e-2
     for mode in range(nm):
         ref = sp.mttkrp_stream(t, mats_c, mode)
         out = mttkrp(cs, mats_g, mode)
-        err = (out.cpu() - ref).abs().max().item()
-        assert err < 1e-8, (cfg, mode, "default", err)
         scale = max(1.0, float(ref.abs().max()))
+        err = (out.cpu() - ref).abs().max().item()
+        assert err < tol * scale, (cfg, mode, "default", err)
         # rows-restricted tiling when supported
         if mttkrp_rows_ok(cs, mode, rank):
             o2 = torch.empty_like(out)
@@ -53,7 +58,7 @@ def one(i):
             d2 = mttkrp(cs, mats_g, mode, deterministic=True)
             assert torch.equal(d1, d2), (cfg, mode, "det-repeat")
             err = (d1.cpu() - ref).abs().max().item()
-            assert err < 1e-8, (cfg, mode, "det", err)
+            assert err < tol * scale, (cfg, mode, "det", err)
     return cfg
 
 
