@@ -50,7 +50,9 @@ def one(i):
             for a, b in zip(cuts, cuts[1:]):
                 mttkrp(cs, mats_g, mode, out=o2, rows=(a, b))
             err = (o2 - out).abs().max().item()
-            assert err < 1e-9 * scale, (cfg, mode, "rows", err)
+            # split-vs-fused summation order differs; fp-level agreement
+            rtol = 1e-9 if dtype == torch.float64 else 1e-3
+            assert err < rtol * scale, (cfg, mode, "rows", err)
         # deterministic path where legal (depth-0 + spec rank + <=5 modes)
         depth = cs.mode_depth[mode]
         if depth == 0 and rank in (4, 8, 16, 32, 64) and nm <= 5:
