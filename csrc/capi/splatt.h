@@ -54,8 +54,8 @@ typedef enum splatt_csf_type {
 typedef enum splatt_tile_type {
   SPLATT_NOTILE = 0,
   SPLATT_DENSETILE = 1,
-  SPLATT_SYNCTILE = 2,   /* deprecated in the reference; accepted, mapped
-                            to the gather-range bucketing analog */
+  SPLATT_SYNCTILE = 2,   /* deprecated in the reference; accepted for
+                            compatibility (the engine chooses tiling) */
   SPLATT_COOPTILE = 3,
 } splatt_tile_type;
 

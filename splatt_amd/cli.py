@@ -52,6 +52,16 @@ def cmd_cpd(args) -> int:
     with TIMERS.time("CPD"):
         if dev == "cpu" and args.native:
             k = sp.cpd_als_cpu_native(t, args.rank, opts)
+        elif dev == "cuda":
+            # production device build: flat streams + LDS-staged buckets
+            # (the same path bench.py and the distributed driver use)
+            from splatt_amd.parallel.dist_cpd import build_shard_csf
+            stage = args.rank if args.rank in (4, 8, 16, 32, 64) else 0
+            cs = build_shard_csf(t.to(dev), list(t.dims), args.csf,
+                                 flat_only=True, stage_rank=stage)
+            print(stats_csf(cs))
+            print(cpd_stats(cs, args.rank, opts))
+            k = sp.cpd_als(cs, args.rank, opts)
         else:
             cs = sp.csf_alloc(t.to(dev), args.csf)
             print(stats_csf(cs))
