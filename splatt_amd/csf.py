@@ -177,6 +177,10 @@ def build_csf(t: SpTensor, perm: Sequence[int], flat_only: bool = False,
             raise ValueError(
                 "device builds support mode dimensions up to 2^31-1 "
                 "(int32 stream labels; shard the mode or build on CPU)")
+        if t.nnz > 0x7FFFFFFF:
+            raise ValueError(
+                "device builds support up to 2^31-1 nonzeros per shard "
+                "(torch sort limit; shard the tensor or build on CPU)")
         if lds_kb <= 0:
             import os
             lds_kb = int(os.environ.get("SPLATT_LDS_KB", "24"))
