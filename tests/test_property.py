@@ -67,3 +67,27 @@ def test_ccp_is_optimal_vs_bruteforce_bound(weights, parts):
     # optimality lower bounds
     assert bn >= max(weights)
     assert bn >= -(-sum(weights) // parts)
+
+
+@settings(max_examples=20, deadline=None)
+@given(data=st.data())
+def test_rows_restricted_mttkrp_tiles_exactly(data):
+    """Any partition of [0, dim) into row ranges reproduces the full
+    MTTKRP exactly (the chunked comm pipeline's invariant)."""
+    from splatt_amd.mttkrp import mttkrp, mttkrp_rows_ok
+    t = rand_tensor(data.draw)
+    t = t.fixed(dedup=True)
+    cs = sp.csf_alloc(t, "all")
+    rank = data.draw(st.sampled_from([4, 8, 16]))
+    mats = [sp.seeded_init(d, rank, m, 3) for m, d in enumerate(t.dims)]
+    mode = data.draw(st.integers(min_value=0, max_value=t.nmodes - 1))
+    assert mttkrp_rows_ok(cs, mode, rank)
+    full = mttkrp(cs, mats, mode)
+    n = t.dims[mode]
+    ncuts = data.draw(st.integers(min_value=0, max_value=4))
+    cuts = sorted({0, n, *(data.draw(st.integers(min_value=0, max_value=n))
+                           for _ in range(ncuts))})
+    out = torch.empty_like(full)
+    for a, b in zip(cuts, cuts[1:]):
+        mttkrp(cs, mats, mode, out=out, rows=(a, b))
+    assert (out - full).abs().max().item() < 1e-12
