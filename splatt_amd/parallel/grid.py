@@ -386,10 +386,10 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
                 owns.append(own_i)
                 rs_works.append(w)
             with tm("SOLVE"):
-                G = torch.ones(F, F, dtype=dtype, device=dev)
-                for o in range(nm):
-                    if o != m:
-                        G *= st.grams[o]
+                # fused Hadamard across the other modes (2 launches
+                # instead of nm-1 in the eager world>1 tail)
+                G = torch.stack([st.grams[o] for o in range(nm)
+                                 if o != m]).prod(dim=0)
                 Ginv = spd_inverse(G)
             with tm("COMM-WAIT"):
                 for w in rs_works:
@@ -452,10 +452,8 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
             with tm("MTTKRP"):
                 mttkrp(st.cs, st.qfactors or st.factors, m, out=mb)
         with tm("SOLVE"):
-            G = torch.ones(F, F, dtype=dtype, device=dev)
-            for o in range(nm):
-                if o != m:
-                    G *= st.grams[o]
+            G = torch.stack([st.grams[o] for o in range(nm)
+                             if o != m]).prod(dim=0)
             Ginv = spd_inverse(G)
         with tm("COMM-WAIT"):
             for w in works:
