@@ -391,12 +391,25 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
                 G = torch.stack([st.grams[o] for o in range(nm)
                                  if o != m]).prod(dim=0)
                 Ginv = spd_inverse(G)
-            with tm("COMM-WAIT"):
-                for w in rs_works:
-                    if w is not None:
-                        w.wait()
+            # per-chunk: as each reduce-scatter lands, solve its rows and
+            # START its all-gather immediately with UNSCALED rows — the
+            # column scale (lambda) is identical on every rank, so
+            # dividing after the gather is algebraically the same update;
+            # this hides the all-gathers under the remaining solves and
+            # the whole lambda/gram section
+            solved, ag = [], []
+            for i in range(C):
+                with tm("COMM-WAIT"):
+                    if rs_works[i] is not None:
+                        rs_works[i].wait()
+                with tm("SOLVE"):
+                    Ai = solve_rows(owns[i], Ginv)
+                solved.append(Ai)
+                with tm("COMM-AG"):
+                    ag.append(_ag_rows_start(Ai, bounds[i + 1] - bounds[i],
+                                             group, gsize))
             own_mb = owns[0] if C == 1 else torch.cat(owns, 0)
-            A_own = solve_rows(own_mb, Ginv)
+            A_own = solved[0] if C == 1 else torch.cat(solved, 0)
             # lambda over GLOBAL rows (owned rows are globally unique)
             if it == 0:
                 s = A_own.square().sum(dim=0)
@@ -408,18 +421,6 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
                 lam = lam.clamp_(min=1.0)
             lam = torch.where(lam == 0, torch.ones_like(lam), lam)
             A_own = A_own / lam
-            A = torch.empty(nrows, F, dtype=dtype, device=dev)
-            # all-gathers start first; gram/fit partials + their small
-            # all-reduces run underneath, then the gathers are drained
-            ag = []
-            with tm("COMM-AG"):
-                ofs = 0
-                for i in range(C):
-                    ni = bounds[i + 1] - bounds[i]
-                    nown = min(ni, (my + 1) * ((ni + gsize - 1) // gsize))                         - min(ni, my * ((ni + gsize - 1) // gsize))
-                    ag.append(_ag_rows_start(A_own[ofs: ofs + nown], ni,
-                                             group, gsize))
-                    ofs += nown
             g = gram(A_own) if A_own.numel() else                 torch.zeros(F, F, dtype=dtype, device=dev)
             _ar(g)
             if m == nm - 1:
@@ -428,10 +429,12 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
                 part = (own_mb.double() * A_own.double()).sum(dim=0)                     if A_own.numel() else torch.zeros(F, dtype=torch.float64,
                                                       device=dev)
                 st._rs_inner = part  # type: ignore[attr-defined]
+            A = torch.empty(nrows, F, dtype=dtype, device=dev)
             with tm("COMM-AG"):
                 for i in range(C):
                     _ag_rows_finish(ag[i], bounds[i + 1] - bounds[i],
                                     A[bounds[i]: bounds[i + 1]])
+            A /= lam
             st.lam = lam
             st.factors[m] = A
             if st.qfactors is not None:
