@@ -241,7 +241,7 @@ def main(argv=None) -> int:
                         "docs/TUNING.md)")
     p.add_argument("--deterministic", action="store_true",
                    help="bitwise-reproducible device CPD (atomic-free "
-                        "kernels, ~60%% throughput; implies --csf all)")
+                        "kernels, >=80%% throughput; implies --csf all)")
     p.add_argument("-v", "--verbose", action="count", default=0)
     p.set_defaults(fn=cmd_cpd)
 
