@@ -101,8 +101,10 @@ def _cmd_cpd_dist(args, world: int) -> int:
     shard = dec.localize(t)
     if use_cuda:
         shard = shard.to(f"cuda:{local_rank % torch.cuda.device_count()}")
+    stage = args.rank if (use_cuda
+                          and args.rank in (4, 8, 16, 32, 64)) else 0
     cs = build_shard_csf(shard, list(t.dims), args.csf,
-                         flat_only=use_cuda)
+                         flat_only=use_cuda, stage_rank=stage)
     opts = sp.CpdOptions(tolerance=args.tol, max_iters=args.its,
                          seed=args.seed, csf_alloc=args.csf)
     k = grid_cpd_als(cs, dec, args.rank, opts)
