@@ -271,3 +271,20 @@ def test_mex_sources_compile(tmp_path):
              "-I" + os.path.join(ROOT, "csrc", "capi"), src],
             capture_output=True, text=True)
         assert r.returncode == 0, (src, r.stderr[-800:])
+
+
+@pytest.mark.gpu
+def test_cli_binary_gpu_engine(tmp_path):
+    """The native `splatt` binary served by the HIP engine end to end
+    (splatt_cpd_als GPU dispatch behind the C CLI)."""
+    t = sp.SpTensor.synthetic([300, 250, 400], 60_000, seed=11).fixed()
+    tns = str(tmp_path / "t.tns")
+    t.save(tns)
+    r = subprocess.run([EXE, "cpd", tns, "-r", "16", "-i", "5", "-t", "0",
+                        "--nowrite"], capture_output=True, text=True,
+                       cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-500:]
+    fit = float(r.stdout.split("Final fit:")[1].split()[0])
+    cs = sp.csf_alloc(t.to("cuda"), "two")
+    ref = sp.cpd_als(cs, 16, sp.CpdOptions(max_iters=5, tolerance=0.0))
+    assert abs(fit - ref.fit) < 1e-4, (fit, ref.fit)
