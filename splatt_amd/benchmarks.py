@@ -63,6 +63,15 @@ def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
             for m, d in enumerate(t.dims)]
     td = t.to(dev)
     cs = csf_alloc(td, policy)
+    # "lds": the production staged/packed device build (what bench.py
+    # and the CPD drivers run) as a benchmarkable algorithm
+    cs_lds = None
+    if dev.type == "cuda" and rank in (4, 8, 16, 32, 64):
+        from splatt_amd.parallel.dist_cpd import build_shard_csf
+        cs_lds = build_shard_csf(td, list(t.dims), policy,
+                                 flat_only=True, stage_rank=rank)
+    elif "lds" in algs:
+        algs.remove("lds")
     gold = None
     if validate:
         gold = [mttkrp_stream(t, [m.cpu() for m in mats], mode)
@@ -89,6 +98,8 @@ def bench_mttkrp(t: SpTensor, rank: int, algs: List[str] = None,
                     return _mttkrp_giga(td, mats, mode)
                 if alg == "ttbox":
                     return _mttkrp_ttbox(td, mats, mode)
+                if alg == "lds":
+                    return mttkrp(cs_lds, mats, mode)
                 return mttkrp(cs, mats, mode, alg=alg, nthreads=nt)
             out = run()  # warmup + result for validation
             if gold is not None:

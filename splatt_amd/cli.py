@@ -274,7 +274,8 @@ def main(argv=None) -> int:
     p = sub.add_parser("bench", help="benchmark MTTKRP algorithms")
     _add_common(p)
     p.add_argument("-r", "--rank", type=int, default=16)
-    p.add_argument("-a", "--algs", default="flat,csf,stream")
+    p.add_argument("-a", "--algs", default="flat,csf,stream",
+                   help="comma list from: flat, csf, stream, giga, ttbox, lds (production staged device path)")
     p.add_argument("-N", "--iters", type=int, default=3)
     p.add_argument("--device", default="auto", choices=["auto", "cpu", "cuda"])
     p.add_argument("--validate", action="store_true")
