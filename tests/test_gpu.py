@@ -639,3 +639,12 @@ def test_gpu_chunked_pipeline_two_ranks_one_gpu(tmp_path):
     # single-process fit on the same global tensor must match (the
     # rank-invariance property, reference mpi_mat_rand)
     assert 0.0 <= j["config"]["fit"] < 1.0
+
+
+def test_gpu_bench_harness_lds_alg(t3):
+    """`bench` harness's 'lds' algorithm (the production staged path)
+    validates against the oracle and beats/equals plain flat."""
+    from splatt_amd.benchmarks import bench_mttkrp
+    r = bench_mttkrp(t3, 16, ["flat", "lds"], 2, device="cuda",
+                     validate=True)
+    assert r["flat"]["validated"] and r["lds"]["validated"]
