@@ -24,10 +24,15 @@ namespace splatt {
 bool capi_gpu_available();
 template <typename V>
 Kruskal<V> cpd_als_gpu(const CsfSet<V> &, int, const Options &);
+template <typename V>
+int mttkrp_gpu(const CsfSet<V> &, void **, int, int, const V * const *,
+               V *);
+void free_dev_csf_cache(void *);
 }
 
 struct splatt_csf {
   CsfSet<double> set;
+  void * dev_cache = nullptr;   // device-resident streams (capi_gpu.cpp)
 };
 
 static Options opts_from_array(const double * o) {
@@ -107,6 +112,8 @@ int splatt_csf_convert(splatt_idx_t nmodes, splatt_idx_t nnz,
 }
 
 void splatt_free_csf(splatt_csf * tensors, const double *) {
+  if (tensors && tensors->dev_cache)
+    free_dev_csf_cache(tensors->dev_cache);
   delete tensors;
 }
 
@@ -162,6 +169,15 @@ int splatt_mttkrp(splatt_idx_t mode, splatt_idx_t ncolumns,
   try {
     (void)options;
     const auto & set = tensors->set;
+    // HIP engine when a GPU is visible (device CSF streams cached on the
+    // handle after the first call); CPU core otherwise or on fallback
+    if (capi_gpu_available()) {
+      auto * h = const_cast<splatt_csf *>(tensors);
+      if (mttkrp_gpu<double>(set, &h->dev_cache, (int)mode, (int)ncolumns,
+                             (double const * const *)matrices,
+                             matout) == 0)
+        return SPLATT_SUCCESS;
+    }
     const auto & c = set.csfs[set.mode_csf[mode]];
     mttkrp_csf_cpu(c, (double const * const *)matrices, matout, (int)mode,
                    (int)ncolumns);

@@ -108,6 +108,89 @@ bool capi_gpu_available() {
   return hipGetDeviceCount(&n) == hipSuccess && n > 0;
 }
 
+// ---- device-resident CSF cache for repeated splatt_mttkrp calls:
+// label streams + values uploaded once per handle, factors per call
+struct DevCsfCache {
+  int ncsf = 0;
+  int nmodes = 0;
+  std::vector<std::vector<DevIBuf>> labs;   // [csf][level]
+  std::vector<DevBuf> vals;                 // [csf]
+};
+
+void free_dev_csf_cache(void * p) {
+  delete reinterpret_cast<DevCsfCache *>(p);
+}
+
+template <typename V>
+int mttkrp_gpu(const CsfSet<V> & set, void ** cache_slot, int mode,
+               int rank, const V * const * mats_host, V * out_host);
+
+template <>
+int mttkrp_gpu<double>(const CsfSet<double> & set, void ** cache_slot,
+                       int mode, int rank,
+                       const double * const * mats_host,
+                       double * out_host) {
+  const Csf<double> & c0 = set.csfs[0];
+  const int nm = c0.nmodes;
+  if (rank > 64) return -1;
+  for (int m = 0; m < nm; ++m)
+    if (c0.dims[m] > 0x7FFFFFFFull) return -1;
+  hipStream_t st = nullptr;
+  auto * cache = reinterpret_cast<DevCsfCache *>(*cache_slot);
+  if (!cache) {
+    cache = new DevCsfCache;
+    cache->ncsf = (int)set.csfs.size();
+    cache->nmodes = nm;
+    cache->labs.resize(cache->ncsf);
+    cache->vals.resize(cache->ncsf);
+    for (int ci = 0; ci < cache->ncsf; ++ci) {
+      const auto & c = set.csfs[ci];
+      cache->labs[ci].resize(nm);
+      for (int l = 0; l < nm; ++l) {
+        auto h = expand_level(c, l);
+        cache->labs[ci][l].p = (int32_t*)dupload(h);
+      }
+      cache->vals[ci].alloc(c.nnz);
+      hip_check(hipMemcpy(cache->vals[ci].p, c.vals.data(),
+                          sizeof(double) * c.nnz, hipMemcpyHostToDevice),
+                "vals H2D");
+    }
+    *cache_slot = cache;
+  }
+  const int ci = set.mode_csf[mode];
+  const auto & c = set.csfs[ci];
+  const int depth = set.mode_depth[mode];
+  std::vector<DevBuf> dmats(nm);
+  for (int m = 0; m < nm; ++m) {
+    dmats[m].alloc(c0.dims[m] * (size_t)rank);
+    hip_check(hipMemcpyAsync(dmats[m].p, mats_host[m],
+                             sizeof(double) * c0.dims[m] * rank,
+                             hipMemcpyHostToDevice, st), "mats H2D");
+  }
+  DevBuf dout;
+  dout.alloc(c0.dims[mode] * (size_t)rank);
+  hip_check(hipMemsetAsync(dout.p, 0,
+                           sizeof(double) * c0.dims[mode] * rank, st),
+            "memset");
+  const int32_t * idxp[8] = {nullptr};
+  const double * matp[8] = {nullptr};
+  int t = 0;
+  for (int l = 0; l < nm; ++l) {
+    if (l == depth) continue;
+    idxp[t] = cache->labs[ci][l].p;
+    matp[t] = dmats[c.dim_perm[l]].p;
+    ++t;
+  }
+  splatt_hip_mttkrp_flat_f64(cache->labs[ci][depth].p, idxp, matp,
+                             cache->vals[ci].p, (int64_t)c.nnz, dout.p,
+                             rank, nm - 1, st);
+  hip_check(hipStreamSynchronize(st), "sync");
+  hip_check(hipMemcpy(out_host, dout.p,
+                      sizeof(double) * c0.dims[mode] * rank,
+                      hipMemcpyDeviceToHost), "out D2H");
+  return 0;
+}
+
 template <typename V>
 Kruskal<V> cpd_als_gpu(const CsfSet<V> & set, int rank, const Options & opts);
 
