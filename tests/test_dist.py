@@ -392,3 +392,20 @@ def test_chunked_comm_pipeline_fit_exact(tmp_path, rsag):
             assert p.exitcode == 0
         fits[chunks] = fit
     assert abs(fits[1] - fits[3]) < 1e-10, fits
+
+
+@pytest.mark.timeout(420)
+def test_dist_fuzz_bounded():
+    """Bounded randomized distributed fuzz (4 trials of
+    scripts/dist_fuzz.py) in CI — random worlds/grids/schedules, fit
+    equal to single-process."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               PYTHONPATH=root + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    r = subprocess.run([sys.executable, "scripts/dist_fuzz.py", "4", "99"],
+                       capture_output=True, text=True, cwd=root, env=env,
+                       timeout=400)
+    assert r.returncode == 0, (r.stdout[-500:], r.stderr[-800:])
+    assert "dist fuzz clean" in r.stdout
