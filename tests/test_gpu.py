@@ -627,11 +627,16 @@ def test_gpu_chunked_pipeline_two_ranks_one_gpu(tmp_path):
                SPLATT_BENCH_CONFIG="small",
                SPLATT_BENCH_BACKEND="gloo",
                SPLATT_FORCE_RS_PRIMS="1")
-    r = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--standalone",
-         "--nnodes=1", "--nproc-per-node", "2", "--local-addr", "127.0.0.1",
-         "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1"],
-        capture_output=True, text=True, cwd=root, env=env, timeout=400)
+    # one retry: torchrun rendezvous on a busy box can transiently fail
+    for attempt in range(2):
+        r = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--standalone",
+             "--nnodes=1", "--nproc-per-node", "2", "--local-addr",
+             "127.0.0.1", "bench.py", "--gpus", "2", "--steps", "2",
+             "--warmup", "1"],
+            capture_output=True, text=True, cwd=root, env=env, timeout=400)
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, r.stderr[-1500:]
     j = json.loads([l for l in r.stdout.strip().splitlines()
                     if l.startswith("{")][0])
