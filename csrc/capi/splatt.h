@@ -86,7 +86,10 @@ typedef enum splatt_option_type {
   SPLATT_OPTION_NOPTIONS
 } splatt_option_type;
 
-/** Opaque CSF tensor handle. */
+/** Opaque CSF tensor handle. Calls taking the same handle are not
+ * thread-safe against each other (the handle lazily caches
+ * device-resident streams on first GPU use); distinct handles are
+ * independent. */
 typedef struct splatt_csf splatt_csf;
 
 /** Kruskal tensor: the CPD output (parity: reference structs.h:25-44). */
