@@ -53,6 +53,21 @@ def one(i):
             # split-vs-fused summation order differs; fp-level agreement
             rtol = 1e-9 if dtype == torch.float64 else 1e-3
             assert err < rtol * scale, (cfg, mode, "rows", err)
+        # reduced-precision factor storage where legal (spec rank, f64)
+        if (dtype == torch.float64 and rank in (4, 8, 16, 32, 64)
+            and rng.random() < 0.3):
+            import os as _os
+            store = rng.choice(["f32", "bf16"])
+            qdt = {"f32": torch.float32,
+                   "bf16": torch.bfloat16}[store]
+            _os.environ["SPLATT_FACTOR_STORE"] = store
+            try:
+                oq = mttkrp(cs, [m.to(qdt) for m in mats_g], mode)
+            finally:
+                del _os.environ["SPLATT_FACTOR_STORE"]
+            stol = 1e-4 if store == "f32" else 5e-2
+            err = (oq.cpu() - ref).abs().max().item()
+            assert err < stol * scale, (cfg, mode, store, err)
         # deterministic path where legal (depth-0 + spec rank + <=5 modes)
         depth = cs.mode_depth[mode]
         if depth == 0 and rank in (4, 8, 16, 32, 64) and nm <= 5:
