@@ -248,7 +248,13 @@ def _gpu_mttkrp_flat(c: Csf, depth: int, mats: List[torch.Tensor],
     key = c.ancestor_expand(depth)
     rank = int(mats[0].shape[1])
     stream = torch.cuda.current_stream().cuda_stream
-    if _use_lds(c, depth, rank):
+    # reduced-precision STORAGE mats need the packed v6 or plain v2
+    # kernels; staged-but-unpacked dispatches (5-mode builds, or
+    # SPLATT_NO_PACK debugging) route to v2
+    storage = mats[0].dtype != c.vals.dtype
+    packed = (getattr(c, "_pack", None) is not None
+              and os.environ.get("SPLATT_NO_PACK") != "1")
+    if _use_lds(c, depth, rank) and (packed or not storage):
         blocks = _stage_blocks(c, rows)
         lvl = blocks["level"]
         if int(blocks["start"].numel()) == 0:
