@@ -17,6 +17,7 @@ import splatt_amd as sp  # noqa: E402
 
 def worker(rank, world, store, q, cfg):
     os.environ["SPLATT_COMM_CHUNKS"] = str(cfg["chunks"])
+    os.environ["SPLATT_COMM_CHUNK_MIN_MB"] = "0"
     if cfg["force_prims"]:
         os.environ["SPLATT_FORCE_RS_PRIMS"] = "1"
     if cfg["no_rsag"]:

@@ -358,7 +358,15 @@ def grid_cpd_step(st: GridCpdState, it: int, overlap: bool = True,
         C = 1
         if distributed and overlap:
             C = _comm_chunks()
+            # pipelining only pays when the exchange is heavy: below
+            # SPLATT_COMM_CHUNK_MIN_MB (default 8) a single collective
+            # beats C x the collective latency
+            import os as _os2
+            min_mb = float(_os2.environ.get("SPLATT_COMM_CHUNK_MIN_MB",
+                                            "8"))
+            payload_mb = nrows * F * st.buf.element_size() / 2**20
             if C > 1 and (nrows < C * dec.repl(m)
+                          or payload_mb < min_mb
                           or not mttkrp_rows_ok(st.cs, m, F)):
                 C = 1
         bounds = [i * nrows // C for i in range(C + 1)]

@@ -61,7 +61,8 @@ def test_bench_driver_argv_world8(tmp_path):
     never the code path."""
     env = dict(os.environ, MASTER_ADDR="127.0.0.1",
                SPLATT_BENCH_CONFIG="small",
-               SPLATT_FORCE_RS_PRIMS="1")
+               SPLATT_FORCE_RS_PRIMS="1",
+               SPLATT_COMM_CHUNK_MIN_MB="0")
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--standalone",
          "--nnodes=1", "--nproc-per-node", "8", "--local-addr", "127.0.0.1",
@@ -83,7 +84,8 @@ def test_bench_driver_argv_world8_4mode(tmp_path):
     driver's exact argv over gloo + forced RS/AG primitives."""
     env = dict(os.environ, MASTER_ADDR="127.0.0.1",
                SPLATT_BENCH_CONFIG="small4",
-               SPLATT_FORCE_RS_PRIMS="1")
+               SPLATT_FORCE_RS_PRIMS="1",
+               SPLATT_COMM_CHUNK_MIN_MB="0")
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--standalone",
          "--nnodes=1", "--nproc-per-node", "8", "--local-addr", "127.0.0.1",

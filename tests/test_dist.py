@@ -344,6 +344,7 @@ def test_grid_cpd_true_rs_primitives(tmp_path, world):
 
 def _chunked_worker(rank, world, file_store, result_q, chunks, rsag):
     os.environ["SPLATT_COMM_CHUNKS"] = str(chunks)
+    os.environ["SPLATT_COMM_CHUNK_MIN_MB"] = "0"
     if rsag:
         os.environ["SPLATT_FORCE_RS_PRIMS"] = "1"
     else:

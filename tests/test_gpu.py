@@ -626,7 +626,8 @@ def test_gpu_chunked_pipeline_two_ranks_one_gpu(tmp_path):
     env = dict(os.environ, MASTER_ADDR="127.0.0.1",
                SPLATT_BENCH_CONFIG="small",
                SPLATT_BENCH_BACKEND="gloo",
-               SPLATT_FORCE_RS_PRIMS="1")
+               SPLATT_FORCE_RS_PRIMS="1",
+               SPLATT_COMM_CHUNK_MIN_MB="0")
     # one retry: torchrun rendezvous on a busy box can transiently fail
     for attempt in range(2):
         r = subprocess.run(
