@@ -161,8 +161,9 @@ mttkrp_flat5_kern(const int32_t * __restrict__ key,
 // S=bf16 with V=double is the documented reduced-precision factor-store
 // mode for HBM-bound shapes: gathered rows shrink 2-4x in cache lines
 // while every multiply-accumulate stays f64 (ROADMAP item 2b).
-template <typename V, int F, int NOTHER, typename S = V, int GBP = 8>
-__global__ void __launch_bounds__(WPB * WAVE)
+template <typename V, int F, int NOTHER, typename S = V, int GBP = 8,
+          int WPB6K = WPB>
+__global__ void __launch_bounds__(WPB6K * WAVE)
 mttkrp_flat6_kern(const int * __restrict__ pack_raw,
                   const S * __restrict__ m0, const S * __restrict__ m1,
                   const S * __restrict__ m2,
@@ -196,11 +197,11 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
       const int nvec = nel / VEC;
       const Vec * src = reinterpret_cast<const Vec *>(m0 + (int64_t)row0 * F);
       Vec * dst = reinterpret_cast<Vec *>(smem);
-      for (int ve = tid; ve < nvec; ve += WPB * WAVE) dst[ve] = src[ve];
-      for (int t = nvec * VEC + tid; t < nel; t += WPB * WAVE)
+      for (int ve = tid; ve < nvec; ve += WPB6K * WAVE) dst[ve] = src[ve];
+      for (int t = nvec * VEC + tid; t < nel; t += WPB6K * WAVE)
         smem[t] = m0[(int64_t)row0 * F + t];
     } else {
-      for (int t = tid; t < nel; t += WPB * WAVE)
+      for (int t = tid; t < nel; t += WPB6K * WAVE)
         smem[t] = m0[(int64_t)row0 * F + t];
     }
   }
@@ -212,7 +213,7 @@ mttkrp_flat6_kern(const int * __restrict__ pack_raw,
   const int c = lane % F;
   const int g = lane / F;
   const int gbase = g * F;
-  const int nsub = WPB * R;
+  const int nsub = WPB6K * R;
   const int sub = wv * R + g;
   const int64_t total = b1 - b0;
   const int64_t gsz = (total + nsub - 1) / nsub;
@@ -269,7 +270,9 @@ void launch_flat6(const int32_t * pack, const S * const mats[3],
                   const int64_t * blk_end, const int32_t * blk_row0,
                   int64_t nblocks, int32_t chunk, int32_t dim0, V * out,
                   int rank, int nother, hipStream_t st) {
-  dim3 grid((uint32_t)nblocks), block(WPB * WAVE);
+  const char * we = getenv("SPLATT_V6_WPB");
+  const int wpb6 = (we && atoi(we) == 8) ? 8 : WPB;
+  dim3 grid((uint32_t)nblocks), block(wpb6 * WAVE);
   const size_t lds = (size_t)chunk * rank * sizeof(S);
   // gather-batch depth A/B lever (register pressure vs loads in flight)
   const char * ge = getenv("SPLATT_V6_GB");
@@ -277,7 +280,9 @@ void launch_flat6(const int32_t * pack, const S * const mats[3],
 #define ARGS6 pack, mats[0], mats[1], mats[2], vals, blk_start, blk_end, \
               blk_row0, chunk, dim0, out
 #define L6(F_, N_) \
-  switch (gb) { \
+  if (wpb6 == 8) { \
+    hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 8, 8>), grid, block, lds, st, ARGS6); \
+  } else switch (gb) { \
     case 4:  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 4>), grid, block, lds, st, ARGS6); break; \
     case 6:  hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 6>), grid, block, lds, st, ARGS6); break; \
     case 12: hipLaunchKernelGGL((mttkrp_flat6_kern<V, F_, N_, S, 12>), grid, block, lds, st, ARGS6); break; \
