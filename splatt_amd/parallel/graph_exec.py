@@ -110,22 +110,27 @@ class GraphStepRunner:
                              * torch.outer(lamd, lamd)).sum()
 
     def capture(self) -> bool:
-        self.capture_error = None
-        try:
-            s = torch.cuda.Stream()
-            with torch.cuda.stream(s):
-                for _ in range(2):   # warm up allocator on side stream
+        # capture can fail transiently (allocator/stream state on a busy
+        # device); one clean retry after a full sync before giving up
+        for attempt in range(2):
+            self.capture_error = None
+            try:
+                s = torch.cuda.Stream()
+                with torch.cuda.stream(s):
+                    for _ in range(2):   # warm up allocator on side stream
+                        self._static_step()
+                torch.cuda.current_stream().wait_stream(s)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
                     self._static_step()
-            torch.cuda.current_stream().wait_stream(s)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                self._static_step()
-            self.graph = g
-            return True
-        except Exception as e:  # noqa: BLE001 - fall back to eager
-            self.capture_error = repr(e)
-            self.graph = None
-            return False
+                self.graph = g
+                return True
+            except Exception as e:  # noqa: BLE001 - fall back to eager
+                self.capture_error = repr(e)
+                self.graph = None
+                if attempt == 0:
+                    torch.cuda.synchronize()
+        return False
 
     def replay(self):
         self.graph.replay()
